@@ -640,7 +640,7 @@ static void g2_mul_u64(g2_t *r, const g2_t *p, uint64_t k) { g2_mul(r, p, &k, 1)
 
 static void g2_clear_cofactor(g2_t *r, const g2_t *p) {
     if (!g_use_fast_cofactor) {
-        g2_mul(r, p, BLS_H2, 7);
+        g2_mul(r, p, BLS_H2, BLS_H2_LIMBS);
         return;
     }
     /* Budroni-Pintore: (z^2-z-1)P + (z-1)psi(P) + psi^2(2P), z = -BLS_U */
