@@ -1,0 +1,31 @@
+"""Build libhbls.so (in-tree, so the .so travels with the repo snapshot).
+
+  python -m harmony_amd.build
+
+hipcc cross-compiles for gfx950 without a GPU present."""
+import os
+import subprocess
+import sys
+
+DIR = os.path.dirname(os.path.abspath(__file__))
+SRC = os.path.join(DIR, "csrc", "hbls_dev.hip")
+OUT = os.path.join(DIR, "libhbls.so")
+
+CMD = [
+    "hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-w",
+    "-shared", "-fPIC", SRC, "-o", OUT,
+]
+
+
+def build(force=False):
+    if not force and os.path.exists(OUT) and \
+            os.path.getmtime(OUT) > os.path.getmtime(SRC):
+        return OUT
+    print("+", " ".join(CMD), flush=True)
+    subprocess.check_call(CMD)
+    return OUT
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)
+    print("built", OUT)

@@ -1,0 +1,224 @@
+"""ctypes binding of libhbls.so — THE PRODUCT PATH (HIP/CDNA4 kernels behind
+the C-ABI in include/hbls.h).
+
+Fails loudly if the library is missing or no AMD GPU is present: there is NO
+CPU fallback here.  The CPU oracle under oracle/ is test infrastructure only.
+"""
+import ctypes
+import os
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "libhbls.so")
+
+HBLS_OK = 1
+HBLS_FALSE = 0
+HBLS_ERR = -1
+HBLS_ERR_NOGPU = -2
+HBLS_ERR_BADINPUT = -3
+
+
+class HblsError(RuntimeError):
+    pass
+
+
+class NoGpuError(HblsError):
+    pass
+
+
+def _load():
+    if not os.path.exists(_SO):
+        raise HblsError(
+            f"libhbls.so not found at {_SO}: build it with "
+            "`python -m harmony_amd.build` (hipcc --offload-arch=gfx950)")
+    lib = ctypes.CDLL(_SO)
+    lib.hbls_version.restype = ctypes.c_char_p
+    lib.hbls_last_kernel_ns.restype = ctypes.c_uint64
+    lib.hbls_committee_build.restype = ctypes.c_void_p
+    lib.hbls_committee_build.argtypes = [ctypes.c_char_p, ctypes.c_size_t]
+    lib.hbls_committee_free.argtypes = [ctypes.c_void_p]
+    lib.hbls_committee_size.argtypes = [ctypes.c_void_p]
+    lib.hbls_committee_size.restype = ctypes.c_size_t
+    lib.hbls_mask_aggregate_g1.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p]
+    lib.hbls_agg_verify.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p,
+                                    ctypes.c_char_p, ctypes.c_size_t]
+    lib.hbls_batch_agg_verify.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p,
+                                          ctypes.c_char_p, ctypes.c_size_t, ctypes.c_size_t,
+                                          ctypes.POINTER(ctypes.c_int32)]
+    lib.hbls_batch_verify_votes.argtypes = [ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32),
+                                            ctypes.c_char_p, ctypes.c_char_p,
+                                            ctypes.c_size_t, ctypes.c_size_t,
+                                            ctypes.POINTER(ctypes.c_int32)]
+    lib.hbls_construct_commit_payload.argtypes = [ctypes.c_uint64, ctypes.c_char_p,
+                                                  ctypes.c_uint64, ctypes.c_int, ctypes.c_char_p]
+    return lib
+
+
+_lib = _load()
+
+
+def _check(rc, what):
+    if rc == HBLS_ERR_NOGPU:
+        raise NoGpuError(f"{what}: no AMD GPU available (product path requires gfx950)")
+    if rc == HBLS_ERR_BADINPUT:
+        raise ValueError(f"{what}: bad input")
+    if rc < 0:
+        raise HblsError(f"{what}: error {rc}")
+    return rc
+
+
+def version() -> str:
+    return _lib.hbls_version().decode()
+
+
+def device_count() -> int:
+    return _lib.hbls_device_count()
+
+
+def init(device=-1):
+    _check(_lib.hbls_init(device), "hbls_init")
+
+
+def last_kernel_ns() -> int:
+    return _lib.hbls_last_kernel_ns()
+
+
+def set_g2_cofactor_mode(fast: bool):
+    _lib.hbls_set_g2_cofactor_mode(int(fast))
+
+
+def pk_from_sk(sk32: bytes) -> bytes:
+    out = ctypes.create_string_buffer(48)
+    _check(_lib.hbls_pk_from_sk(sk32, out), "pk_from_sk")
+    return out.raw
+
+
+def batch_pk_from_sk(sks: bytes, batch: int) -> bytes:
+    out = ctypes.create_string_buffer(48 * batch)
+    _check(_lib.hbls_batch_pk_from_sk(sks, batch, out), "batch_pk_from_sk")
+    return out.raw
+
+
+def sign_hash(sk32: bytes, msg: bytes) -> bytes:
+    out = ctypes.create_string_buffer(96)
+    _check(_lib.hbls_sign_hash(sk32, msg, len(msg), out), "sign_hash")
+    return out.raw
+
+
+def batch_sign(sks: bytes, msgs: bytes, mlen: int, batch: int) -> bytes:
+    out = ctypes.create_string_buffer(96 * batch)
+    _check(_lib.hbls_batch_sign(sks, msgs, mlen, batch, out), "batch_sign")
+    return out.raw
+
+
+def hash_to_g2(msg: bytes) -> bytes:
+    out = ctypes.create_string_buffer(96)
+    _check(_lib.hbls_hash_to_g2(msg, len(msg), out), "hash_to_g2")
+    return out.raw
+
+
+def batch_hash_to_g2(msgs: bytes, mlen: int, batch: int) -> bytes:
+    out = ctypes.create_string_buffer(96 * batch)
+    _check(_lib.hbls_batch_hash_to_g2(msgs, mlen, batch, out), "batch_hash_to_g2")
+    return out.raw
+
+
+def verify_hash(pk48: bytes, sig96: bytes, msg: bytes) -> bool:
+    rc = _lib.hbls_verify_hash(pk48, sig96, msg, len(msg))
+    return _check(rc, "verify_hash") == HBLS_OK
+
+
+def g1_add(a: bytes, b: bytes) -> bytes:
+    out = ctypes.create_string_buffer(48)
+    _check(_lib.hbls_g1_add(a, b, out), "g1_add")
+    return out.raw
+
+
+def g1_sub(a: bytes, b: bytes) -> bytes:
+    out = ctypes.create_string_buffer(48)
+    _check(_lib.hbls_g1_sub(a, b, out), "g1_sub")
+    return out.raw
+
+
+def g2_add(a: bytes, b: bytes) -> bytes:
+    out = ctypes.create_string_buffer(96)
+    _check(_lib.hbls_g2_add(a, b, out), "g2_add")
+    return out.raw
+
+
+def g1_check(p48: bytes) -> bool:
+    return _check(_lib.hbls_g1_check(p48), "g1_check") == HBLS_OK
+
+
+def g2_check(p96: bytes) -> bool:
+    return _check(_lib.hbls_g2_check(p96), "g2_check") == HBLS_OK
+
+
+def msm_g1(points: bytes, scalars: bytes, n: int) -> bytes:
+    out = ctypes.create_string_buffer(48)
+    _check(_lib.hbls_msm_g1(points, scalars, n, out), "msm_g1")
+    return out.raw
+
+
+def batch_keccak256(msgs: bytes, mlen: int, batch: int) -> bytes:
+    out = ctypes.create_string_buffer(32 * batch)
+    _check(_lib.hbls_batch_keccak256(msgs, mlen, batch, out), "batch_keccak")
+    return out.raw
+
+
+def construct_commit_payload(block_num: int, hash32: bytes, view_id: int,
+                             staking: bool = True) -> bytes:
+    out = ctypes.create_string_buffer(48)
+    n = _lib.hbls_construct_commit_payload(block_num, hash32, view_id, int(staking), out)
+    return out.raw[:n]
+
+
+def parse_commit_sig_bitmap(payload: bytes):
+    sig = ctypes.create_string_buffer(96)
+    bm = ctypes.create_string_buffer(max(len(payload), 96) - 96 + 1)
+    n = _lib.hbls_parse_commit_sig_bitmap(payload, len(payload), sig, bm, len(payload))
+    if n < 0:
+        raise ValueError("payload too short")
+    return sig.raw, bm.raw[:n]
+
+
+class Committee:
+    """Device-resident, validated pubkey table (UpdateParticipants equivalent)."""
+
+    def __init__(self, pks_cat: bytes, n: int):
+        self._h = _lib.hbls_committee_build(pks_cat, n)
+        if not self._h:
+            if device_count() == 0:
+                raise NoGpuError("committee_build: no AMD GPU")
+            raise ValueError("committee_build: invalid pubkey in table")
+        self.n = n
+
+    def mask_aggregate(self, bitmap: bytes) -> bytes:
+        out = ctypes.create_string_buffer(48)
+        _check(_lib.hbls_mask_aggregate_g1(self._h, bitmap, out), "mask_aggregate")
+        return out.raw
+
+    def agg_verify(self, bitmap: bytes, sig96: bytes, msg: bytes) -> bool:
+        rc = _lib.hbls_agg_verify(self._h, bitmap, sig96, msg, len(msg))
+        return _check(rc, "agg_verify") == HBLS_OK
+
+    def batch_agg_verify(self, bitmaps: bytes, sigs: bytes, msgs: bytes,
+                         mlen: int, batch: int):
+        res = (ctypes.c_int32 * batch)()
+        _check(_lib.hbls_batch_agg_verify(self._h, bitmaps, sigs, msgs, mlen,
+                                          batch, res), "batch_agg_verify")
+        return list(res)
+
+    def batch_verify_votes(self, key_idx, sigs: bytes, msgs: bytes, mlen: int):
+        batch = len(key_idx)
+        idx = (ctypes.c_uint32 * batch)(*key_idx)
+        res = (ctypes.c_int32 * batch)()
+        _check(_lib.hbls_batch_verify_votes(self._h, idx, sigs, msgs, mlen,
+                                            batch, res), "batch_verify_votes")
+        return list(res)
+
+    def __del__(self):
+        try:
+            if getattr(self, "_h", None):
+                _lib.hbls_committee_free(self._h)
+        except Exception:
+            pass

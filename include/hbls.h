@@ -1,0 +1,161 @@
+/* hbls.h — C-ABI of the MI355X-native BLS12-381 library (libhbls.so).
+ *
+ * This is the drop-in boundary for harmony-one/harmony's BLS FFI: the Go repo
+ * reaches all curve/pairing arithmetic through the cgo package
+ * `github.com/harmony-one/bls/ffi/go/bls` (imported as bls_core at e.g.
+ * crypto/bls/bls.go:8, consensus/quorum/quorum.go:11, internal/chain/sig.go:6).
+ * Each entry point below names the herumi bls C function (bls/bls.h /
+ * src/bls_c_impl.hpp) the Go FFI binds and which this library replaces.
+ * A maintainer wires this in with a cgo shim — see INTEGRATION.md.
+ *
+ * Conventions (match the herumi FFI the reference uses):
+ *   - curve: BLS12-381 with BLS_SWAP_G=1 (reference Makefile:71-73):
+ *     public keys in G1 (48 B compressed), signatures in G2 (96 B).
+ *   - all points cross the ABI in herumi little-endian compressed form
+ *     (or as opaque device-resident handles for committee tables).
+ *   - secret keys: 32 B little-endian Fr, value < r.
+ *   - return 0 = failure/reject, 1 = success/accept, negative = error,
+ *     unless noted.  Thread-safe after hbls_init() (no mutable globals).
+ *   - REQUIRES an AMD GPU (gfx950).  There is no CPU fallback: calls fail
+ *     loudly (HBLS_ERR_NOGPU) when no device is present.
+ */
+#ifndef HBLS_H
+#define HBLS_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+enum {
+    HBLS_OK = 1,
+    HBLS_FALSE = 0,
+    HBLS_ERR = -1,
+    HBLS_ERR_NOGPU = -2,
+    HBLS_ERR_BADINPUT = -3,
+};
+
+/* blsInit(BLS12_381) (bls.h) — called once per process (crypto/bls/mask.go:18-20).
+ * Initializes the HIP device context; device = HIP device index (normally
+ * LOCAL_RANK; pass -1 for current device). */
+int hbls_init(int device);
+
+/* number of usable GPUs (0 -> every call returns HBLS_ERR_NOGPU) */
+int hbls_device_count(void);
+
+/* ---- scalar drop-ins (one call per object, like the cgo surface) ---- */
+
+/* blsGetPublicKey: pk = sk * basePoint (SecretKey.GetPublicKey) */
+int hbls_pk_from_sk(const uint8_t sk32[32], uint8_t pk48[48]);
+
+/* blsSignHash: sig = sk * H2(msg)  (SecretKey.SignHash; construct.go:101,110) */
+int hbls_sign_hash(const uint8_t sk32[32], const uint8_t *msg, size_t msg_len,
+                   uint8_t sig96[96]);
+
+/* blsVerifyHash: e(pub, H2(msg)) == e(base, sig)  (Sign.VerifyHash;
+ * leader.go:173,287, validator.go:228, engine.go:638).
+ * Returns HBLS_OK accept / HBLS_FALSE reject / HBLS_ERR_BADINPUT. */
+int hbls_verify_hash(const uint8_t pk48[48], const uint8_t sig96[96],
+                     const uint8_t *msg, size_t msg_len);
+
+/* blsPublicKeyAdd / blsPublicKeySub (PublicKey.Add/Sub; mask.go:126-130).
+ * Zero input (48 zero bytes) is the identity, matching zero-value structs. */
+int hbls_g1_add(const uint8_t a48[48], const uint8_t b48[48], uint8_t out48[48]);
+int hbls_g1_sub(const uint8_t a48[48], const uint8_t b48[48], uint8_t out48[48]);
+
+/* blsSignatureAdd (Sign.Add; mask.go:57-64, construct.go:99-105) */
+int hbls_g2_add(const uint8_t a96[96], const uint8_t b96[96], uint8_t out96[96]);
+
+/* blsPublicKeyDeserialize / blsSignatureDeserialize validity check
+ * (subgroup-checked, as herumi does on deserialize) */
+int hbls_g1_check(const uint8_t p48[48]);
+int hbls_g2_check(const uint8_t p96[96]);
+
+/* blsHashToSignature equivalent: H2(msg) serialized (used by tests) */
+int hbls_hash_to_g2(const uint8_t *msg, size_t msg_len, uint8_t out96[96]);
+
+/* mcl G2 cofactor mode: 1 = Budroni-Pintore fast (default), 0 = full h2.
+ * Mirrors mcl's useOriginalG2cofactor_ switch; see DESIGN.md parity notes. */
+void hbls_set_g2_cofactor_mode(int fast);
+
+/* ---- committee table (device-resident pubkey table) ----
+ * Mirrors Decider.UpdateParticipants (consensus/quorum/quorum.go:326-334) +
+ * the deserialized-pubkey cache (crypto/bls/bls.go:30-33, mask.go:13-15):
+ * upload once per epoch, aggregate/verify against it many times. */
+typedef struct hbls_committee hbls_committee_t;
+
+/* build from n concatenated 48-B compressed keys; validates every key
+ * (curve + subgroup) on the GPU; returns NULL on any invalid key */
+hbls_committee_t *hbls_committee_build(const uint8_t *pks48, size_t n);
+void hbls_committee_free(hbls_committee_t *c);
+size_t hbls_committee_size(const hbls_committee_t *c);
+
+/* Mask.SetMask / masked aggregate (crypto/bls/mask.go:113-134):
+ * AggregatePublic = sum of pk_i with bit i set.  bitmap is ceil(n/8) bytes,
+ * little-endian bit order (bit i of byte i/8 = cosigner i). */
+int hbls_mask_aggregate_g1(const hbls_committee_t *c, const uint8_t *bitmap,
+                           uint8_t out48[48]);
+
+/* one aggregate verify: DecodeSigBitmap + IsQuorumAchievedByMask's cryptoleg +
+ * VerifyHash (internal/chain/sig.go:37-49 -> engine.go:630-642) */
+int hbls_agg_verify(const hbls_committee_t *c, const uint8_t *bitmap,
+                    const uint8_t sig96[96], const uint8_t *msg, size_t msg_len);
+
+/* ---- batch entry points (the GPU-native surface) ---- */
+
+/* batch of independent aggregate-verifies against one committee:
+ * bitmaps: batch * ceil(n/8) bytes; sigs: batch * 96; msgs: batch * msg_len.
+ * results[j] = HBLS_OK / HBLS_FALSE / HBLS_ERR_BADINPUT. */
+int hbls_batch_agg_verify(const hbls_committee_t *c, const uint8_t *bitmaps,
+                          const uint8_t *sigs96, const uint8_t *msgs,
+                          size_t msg_len, size_t batch, int32_t *results);
+
+/* batch verify of per-signer votes (leader's onCommit loop, leader.go:221-301):
+ * item j checks sig[j] by committee member key_idx[j] on msgs[j]. */
+int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t *key_idx,
+                            const uint8_t *sigs96, const uint8_t *msgs,
+                            size_t msg_len, size_t batch, int32_t *results);
+
+/* batch hash-to-G2 (toG / mcl mapToG2 legacy) */
+int hbls_batch_hash_to_g2(const uint8_t *msgs, size_t msg_len, size_t batch,
+                          uint8_t *out96s);
+
+/* batch sign: sigs[j] = sk[j] * H2(msgs[j]) */
+int hbls_batch_sign(const uint8_t *sks32, const uint8_t *msgs, size_t msg_len,
+                    size_t batch, uint8_t *sigs96);
+
+/* batch sk->pk */
+int hbls_batch_pk_from_sk(const uint8_t *sks32, size_t batch, uint8_t *pks48);
+
+/* G1 MSM: out = sum_i scalar_i * P_i (scalars 32B LE each; general building
+ * block behind mask aggregation; Pippenger on device) */
+int hbls_msm_g1(const uint8_t *points48, const uint8_t *scalars32, size_t n,
+                uint8_t out48[48]);
+
+/* ConstructCommitPayload (consensus/signature/signature.go:12-24); returns 40/48 */
+int hbls_construct_commit_payload(uint64_t block_num, const uint8_t hash32[32],
+                                  uint64_t view_id, int staking, uint8_t out[48]);
+
+/* ParseCommitSigAndBitmap (internal/chain/sig.go:22-35); returns bitmap length */
+int hbls_parse_commit_sig_bitmap(const uint8_t *payload, size_t len,
+                                 uint8_t sig96[96], uint8_t *bitmap, size_t bitmap_cap);
+
+/* batch Keccak-256 (consensus message digests, crypto/hash/hash.go:9-15) */
+int hbls_batch_keccak256(const uint8_t *msgs, size_t msg_len, size_t batch,
+                         uint8_t *out32s);
+
+/* ---- introspection ---- */
+const char *hbls_version(void);
+/* measured device 64x64->128 integer-mad throughput (ops/s) — the VALU
+ * roofline peak for this integer-bound path (no datasheet figure exists) */
+double hbls_mad_peak_ops(void);
+/* elapsed device-time of the last batch call on this thread's stream, in ns
+ * (HIP events); 0 if unavailable.  For bench.py's roofline leg. */
+uint64_t hbls_last_kernel_ns(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* HBLS_H */

@@ -162,3 +162,16 @@ def set_g2_cofactor_mode(fast: bool):
 
 def nthreads() -> int:
     return _lib.oracle_nthreads()
+
+
+_lib.oracle_op_count.restype = ctypes.c_uint64
+
+
+def reset_op_count():
+    _lib.oracle_reset_op_count()
+
+
+def op_count() -> int:
+    """Fp-multiplications executed on THIS thread since the last reset —
+    the algorithmic-work meter for roofline accounting."""
+    return _lib.oracle_op_count()

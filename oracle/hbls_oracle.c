@@ -92,8 +92,16 @@ static void fp_neg(fp_t *r, const fp_t *x) {
 }
 static void fp_dbl(fp_t *r, const fp_t *x) { fp_add(r, x, x); }
 
+/* algorithmic-work counter: every Fp multiplication (the unit all cost
+ * estimates in SURVEY.md §8d are stated in).  Thread-local; used by bench.py
+ * to derive the EXACT algorithmic op count of one aggregate-verify. */
+static __thread uint64_t g_fp_mul_count = 0;
+void oracle_reset_op_count(void) { g_fp_mul_count = 0; }
+uint64_t oracle_op_count(void) { return g_fp_mul_count; }
+
 /* CIOS Montgomery multiplication */
 static void fp_mul(fp_t *r, const fp_t *x, const fp_t *y) {
+    g_fp_mul_count++;
     uint64_t t[8];
     memset(t, 0, sizeof(t));
     for (int i = 0; i < 6; i++) {
