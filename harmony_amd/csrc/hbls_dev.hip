@@ -951,10 +951,15 @@ DEVN int verify_pairing(const g1_t &pub, const g2_t &hm, const g2aff_t &sig_aff,
       ba.x = nb.x; ba.y = nb.y; }
     g2_t hmc = hm;
     g2_to_affine(ha, hmc);
-    fp12_t f;
+    /* the Miller accumulator's squarings require f to start at 1, so the two
+     * loops run separately and multiply (shared-squaring fusion is a later
+     * optimization) */
+    fp12_t f, f2;
     fp12_one(f);
     miller_loop_acc(f, ha, pa);
-    miller_loop_acc(f, sig_aff, ba);
+    fp12_one(f2);
+    miller_loop_acc(f2, sig_aff, ba);
+    fp12_mul(f, f, f2);
     fp12_conj(f, f);
     final_exp(f, f);
     return fp12_is_one(f) ? 1 : 0;
