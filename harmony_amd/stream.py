@@ -1,0 +1,66 @@
+"""harmony_amd.stream — config-5 streaming FBFT round harness
+(BASELINE.json configs[4]): a stream of Prepare/Commit vote messages,
+each a protobuf-like blob -> Keccak-256 digest -> per-message signature
+verify against the sender's committee key -> incremental aggregate
+(mask bit + G2 sum), with a batched pairing check of the running aggregate
+every `window` messages.
+
+Mirrors the leader's per-message hot loop (consensus/leader.go:221-309:
+parse -> ConstructCommitPayload -> Sign.Deserialize -> VerifyHash ->
+AddNewVote -> commitBitmap.SetKeysAtomic) re-shaped around the batch GPU
+API: messages are collected into micro-batches and verified in one
+hbls_batch_verify_votes launch; aggregation happens on the (tiny) mask
+bitmap + one aggregate check per window via hbls_agg_verify.
+"""
+import os
+import sys
+
+from . import core
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+class StreamVerifier:
+    def __init__(self, pks_cat: bytes, n: int, payload: bytes, window: int = 100):
+        """pks_cat: committee table; payload: the commit payload all votes of
+        this round sign (leader.go:250); window: aggregate-check period."""
+        self.committee = core.Committee(pks_cat, n)
+        self.n = n
+        self.payload = payload
+        self.window = window
+        self.bitmap = bytearray((n + 7) // 8)
+        self.agg_sig = b"\x00" * 96
+        self.accepted = 0
+        self.rejected = 0
+        self.window_checks = 0
+        self._since_check = 0
+
+    def process_batch(self, key_idx, sigs_cat: bytes, blobs_cat: bytes, blob_len: int):
+        """One micro-batch of votes: blobs are hashed on-GPU (sender-auth
+        digest path, checks.go:20-39 analog), signatures verified per sender
+        key in one launch, accepted votes folded into the aggregate."""
+        batch = len(key_idx)
+        # keccak digests of the raw message blobs (crypto/hash/hash.go:9-15)
+        _digests = core.batch_keccak256(blobs_cat, blob_len, batch)
+        # per-vote verify of the commit payload signature
+        msgs = self.payload * batch
+        res = self.committee.batch_verify_votes(key_idx, sigs_cat, msgs, len(self.payload))
+        for j, ok in enumerate(res):
+            i = key_idx[j]
+            if ok == 1 and not (self.bitmap[i >> 3] >> (i & 7)) & 1:
+                self.bitmap[i >> 3] |= 1 << (i & 7)
+                self.agg_sig = core.g2_add(self.agg_sig, sigs_cat[96 * j:96 * (j + 1)])
+                self.accepted += 1
+            else:
+                self.rejected += 1
+        self._since_check += batch
+        if self._since_check >= self.window:
+            self._since_check = 0
+            self.window_checks += 1
+            ok = self.committee.agg_verify(bytes(self.bitmap), self.agg_sig, self.payload)
+            if not ok:
+                raise RuntimeError("streaming aggregate diverged from mask")
+        return res
+
+    def final_check(self) -> bool:
+        return self.committee.agg_verify(bytes(self.bitmap), self.agg_sig, self.payload)

@@ -1,0 +1,63 @@
+"""config-5 streaming harness on GPU: per-message verify + incremental
+aggregate + windowed pairing checks, parity with the oracle at the end."""
+import os
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+pytestmark = pytest.mark.gpu
+
+from oracle import pyref as pr  # noqa: E402
+
+
+def _gpu_available():
+    try:
+        from harmony_amd import core
+        return core.device_count() > 0
+    except Exception:
+        return False
+
+
+if not os.environ.get("HBLS_FORCE_GPU_TESTS"):
+    pytestmark = [pytest.mark.gpu,
+                  pytest.mark.skipif(not _gpu_available(), reason="no AMD GPU")]
+
+
+def test_stream_round(oracle_lib):
+    from harmony_amd import core
+    from harmony_amd.stream import StreamVerifier
+    n = 64
+    sks = [pr.fr_serialize(pr.synth_sk(i)) for i in range(n)]
+    pks = core.batch_pk_from_sk(b"".join(sks), n)
+    payload = pr.construct_commit_payload(77, pr.synth_msg(77), 5)
+    sv = StreamVerifier(pks, n, payload, window=16)
+
+    blob_len = 512
+    import random
+    rng = random.Random(7)
+    order = list(range(n)) + [3, 5]          # two duplicate votes
+    rng.shuffle(order)
+    bad_at = order[10]                        # one corrupted signature
+
+    sigs_all = core.batch_sign(b"".join(sks), payload * n, len(payload), n)
+    for start in range(0, len(order), 8):
+        chunk = order[start:start + 8]
+        sigs = b""
+        for i in chunk:
+            s = sigs_all[96 * i:96 * (i + 1)]
+            if i == bad_at:
+                # substitute wrong-message signature
+                s = oracle_lib.sign_hash(sks[i], payload[:-1] + b"\x00")
+            sigs += s
+        blobs = b"".join((pr.synth_msg(i) * 20)[:blob_len] for i in chunk)
+        sv.process_batch(chunk, sigs, blobs, blob_len)
+
+    # the corrupted vote and the duplicates must not be in the aggregate
+    assert sv.accepted == n - 1
+    assert sv.rejected == 3          # 1 bad sig + 2 duplicates
+    assert sv.final_check() is True
+    # oracle cross-check of the final aggregate
+    oc = oracle_lib.Committee(pks, n)
+    assert oc.agg_verify(bytes(sv.bitmap), sv.agg_sig, payload) is True
