@@ -33,6 +33,8 @@ def _load():
     lib = ctypes.CDLL(_SO)
     lib.hbls_version.restype = ctypes.c_char_p
     lib.hbls_last_kernel_ns.restype = ctypes.c_uint64
+    lib.hbls_last_stage_ns.restype = ctypes.c_uint64
+    lib.hbls_mad_peak_ops.restype = ctypes.c_double
     lib.hbls_committee_build.restype = ctypes.c_void_p
     lib.hbls_committee_build.argtypes = [ctypes.c_char_p, ctypes.c_size_t]
     lib.hbls_committee_free.argtypes = [ctypes.c_void_p]

@@ -115,7 +115,7 @@ DEV void fp_neg(fp_t &r, const fp_t &x) {
 }
 DEV void fp_dbl(fp_t &r, const fp_t &x) { fp_add(r, x, x); }
 
-DEVN void fp_mul(fp_t &r, const fp_t &x, const fp_t &y) {
+DEV void fp_mul(fp_t &r, const fp_t &x, const fp_t &y) {
     uint64_t t[7];
 #pragma unroll
     for (int i = 0; i < 7; i++) t[i] = 0;
@@ -165,7 +165,7 @@ DEV void fp_to_mont(fp_t &r, const uint64_t raw[6]) {
     for (int i = 0; i < 6; i++) { t.l[i] = raw[i]; r2.l[i] = BLS_R2P[i]; }
     fp_mul(r, t, r2);
 }
-DEVN void fp_from_mont(uint64_t raw[6], const fp_t &x) {
+DEV void fp_from_mont(uint64_t raw[6], const fp_t &x) {
     fp_t one, t;
     fp_zero(one);
     one.l[0] = 1;
@@ -218,7 +218,7 @@ DEV void fp2_sub(fp2_t &r, const fp2_t &x, const fp2_t &y) { fp_sub(r.a, x.a, y.
 DEV void fp2_neg(fp2_t &r, const fp2_t &x) { fp_neg(r.a, x.a); fp_neg(r.b, x.b); }
 DEV void fp2_conj(fp2_t &r, const fp2_t &x) { r.a = x.a; fp_neg(r.b, x.b); }
 DEV void fp2_dbl(fp2_t &r, const fp2_t &x) { fp2_add(r, x, x); }
-DEVN void fp2_mul(fp2_t &r, const fp2_t &x, const fp2_t &y) {
+DEV void fp2_mul(fp2_t &r, const fp2_t &x, const fp2_t &y) {
     fp_t ac, bd, ab, cd, t;
     fp_mul(ac, x.a, y.a);
     fp_mul(bd, x.b, y.b);
@@ -230,7 +230,7 @@ DEVN void fp2_mul(fp2_t &r, const fp2_t &x, const fp2_t &y) {
     fp_sub(r.a, ac, bd);
     r.b = t;
 }
-DEVN void fp2_sqr(fp2_t &r, const fp2_t &x) {
+DEV void fp2_sqr(fp2_t &r, const fp2_t &x) {
     fp_t s, d, m;
     fp_add(s, x.a, x.b);
     fp_sub(d, x.a, x.b);
@@ -485,19 +485,19 @@ DEVN void g2_mul(g2_t &r, const g2_t &p, const uint64_t *k, int n) {
     r = acc;
 }
 DEVN void g2_psi(g2_t &r, const g2_t &p) {
+    /* psi on Jacobian coords directly: x=X/Z^2, y=Y/Z^3 ->
+     * (cx*conj(X), cy*conj(Y), conj(Z)) — no inversion needed. */
     if (g2_is_inf(p)) { r = p; return; }
-    g2aff_t a, pa;
-    g2_to_affine(a, p);
     fp2_t cx, cy, t;
     fp_load(cx.a, BLS_PSI_CX_A); fp_load(cx.b, BLS_PSI_CX_B);
     fp_load(cy.a, BLS_PSI_CY_A); fp_load(cy.b, BLS_PSI_CY_B);
-    fp2_conj(t, a.x); fp2_mul(pa.x, t, cx);
-    fp2_conj(t, a.y); fp2_mul(pa.y, t, cy);
-    g2_from_affine(r, pa);
+    fp2_conj(t, p.x); fp2_mul(r.x, t, cx);
+    fp2_conj(t, p.y); fp2_mul(r.y, t, cy);
+    fp2_conj(r.z, p.z);
 }
 
 /* ================================================================ serialization */
-DEVN void fp_to_le48(uint8_t out[48], const fp_t &x) {
+DEV void fp_to_le48(uint8_t out[48], const fp_t &x) {
     uint64_t raw[6];
     fp_from_mont(raw, x);
 #pragma unroll
@@ -506,7 +506,7 @@ DEVN void fp_to_le48(uint8_t out[48], const fp_t &x) {
         for (int j = 0; j < 8; j++)
             out[i * 8 + j] = (uint8_t)(raw[i] >> (8 * j));
 }
-DEVN bool fp_from_le48(fp_t &x, const uint8_t in[48]) {
+DEV bool fp_from_le48(fp_t &x, const uint8_t in[48]) {
     uint64_t raw[6];
 #pragma unroll
     for (int i = 0; i < 6; i++) {
@@ -561,6 +561,29 @@ DEV bool g2_in_subgroup(const g2_t &p) {
     g2_mul(t, p, BLS_R, 4);
     return g2_is_inf(t);
 }
+DEV bool g2_eq_jac(const g2_t &p, const g2_t &q) {
+    if (g2_is_inf(p) || g2_is_inf(q)) return g2_is_inf(p) && g2_is_inf(q);
+    fp2_t z1z1, z2z2, a, b;
+    fp2_sqr(z1z1, p.z); fp2_sqr(z2z2, q.z);
+    fp2_mul(a, p.x, z2z2); fp2_mul(b, q.x, z1z1);
+    if (!fp2_eq(a, b)) return false;
+    fp2_mul(a, p.y, q.z); fp2_mul(a, a, z2z2);
+    fp2_mul(b, q.y, p.z); fp2_mul(b, b, z1z1);
+    return fp2_eq(a, b);
+}
+/* Scott's G2 membership criterion for BLS12-381: Q in E'(Fp2) is in the
+ * order-r subgroup iff psi(Q) == [z]Q  (z negative: [z]Q = -[|z|]Q).
+ * ~4x cheaper than the full [r]Q == inf check; equivalence property-tested
+ * against the r-mult on valid and out-of-subgroup points (GPU tests). */
+DEVN bool g2_in_subgroup_fast(const g2_t &p) {
+    if (g2_is_inf(p)) return true;
+    g2_t lhs, rhs;
+    g2_psi(lhs, p);
+    uint64_t u = BLS_U;
+    g2_mul(rhs, p, &u, 1);
+    g2_neg(rhs, rhs);
+    return g2_eq_jac(lhs, rhs);
+}
 /* 1 ok; 0 bad; infinity accepted (idx==inf flag via out z=0) */
 DEVN bool g1_deserialize(g1_t &p, const uint8_t in[48], bool check_subgroup) {
     if (bytes_all_zero(in, 48)) { g1_set_inf(p); return true; }
@@ -596,7 +619,7 @@ DEVN bool g2_deserialize(g2_t &p, const uint8_t in[96], bool check_subgroup) {
     if (!fp2_sqrt(a.y, y2)) return false;
     if (fp2_is_odd(a.y) != odd) fp2_neg(a.y, a.y);
     g2_from_affine(p, a);
-    if (check_subgroup && !g2_in_subgroup(p)) return false;
+    if (check_subgroup && !g2_in_subgroup_fast(p)) return false;
     return true;
 }
 
@@ -1202,6 +1225,14 @@ __global__ void k_g2_check(const uint8_t *p96, int32_t *ok) {
     g2_t p;
     *ok = g2_deserialize(p, p96, true) ? 1 : 0;
 }
+/* test support: decompress WITHOUT subgroup check, then report both
+ * membership methods (full [r]Q and psi-criterion) for equivalence tests */
+__global__ void k_g2_subgroup_methods(const uint8_t *p96, int32_t *out2) {
+    g2_t p;
+    if (!g2_deserialize(p, p96, false)) { out2[0] = out2[1] = -1; return; }
+    out2[0] = g2_in_subgroup(p) ? 1 : 0;
+    out2[1] = g2_in_subgroup_fast(p) ? 1 : 0;
+}
 
 /* MSM v1: blocks of 256 threads; thread t of block b handles point b*256+t,
  * does full scalar mult, LDS tree reduce; second kernel reduces block results. */
@@ -1609,6 +1640,18 @@ extern "C" int hbls_g1_check(const uint8_t p48[48]) {
     int32_t ok;
     HIP_OK(hipMemcpy(&ok, dok.p, 4, hipMemcpyDeviceToHost));
     return ok ? HBLS_OK : HBLS_FALSE;
+}
+extern "C" int hbls_g2_subgroup_methods(const uint8_t p96[96], int32_t out2[2]) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    DevBuf dp(96), dout(8);
+    if (dp.err || dout.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(dp.p, p96, 96, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_g2_subgroup_methods, dim3(1), dim3(1), 0, 0,
+                       dp.as<uint8_t>(), dout.as<int32_t>());
+    HIP_OK(hipDeviceSynchronize());
+    HIP_OK(hipMemcpy(out2, dout.p, 8, hipMemcpyDeviceToHost));
+    return HBLS_OK;
 }
 extern "C" int hbls_g2_check(const uint8_t p96[96]) {
     int rc = require_gpu();

@@ -207,6 +207,29 @@ def test_batch_verify_votes(core, capi, keys16):
     assert gc.batch_verify_votes(idx, sigs2, msg * 3, len(msg)) == [1, 0, 1]
 
 
+def test_g2_subgroup_method_equivalence(core, capi):
+    """psi-criterion membership == full [r]Q membership, on an in-subgroup
+    signature and on an out-of-subgroup curve point."""
+    import ctypes
+    out = (ctypes.c_int32 * 2)()
+    sig = capi.sign_hash(sk_bytes(0), pr.synth_msg(0))
+    core._lib.hbls_g2_subgroup_methods(sig, out)
+    assert list(out) == [1, 1]
+    # out-of-subgroup point on E'(Fp2): sweep x = (c, 1) until y^2 is a QR
+    c = 1
+    while True:
+        x = (c, 1)
+        y2 = pr.f2_add(pr.f2_mul(pr.f2_sqr(x), x), pr.B2)
+        y = pr.f2_sqrt(y2)
+        if y is not None and pr.g2_mul((x, y), pr.R) is not None:
+            bad = pr.g2_serialize((x, y))
+            break
+        c += 1
+    core._lib.hbls_g2_subgroup_methods(bad, out)
+    assert list(out) == [0, 0]
+    assert not core.g2_check(bad)
+
+
 def test_msm_parity_small(core):
     n = 4
     pts = [pr.get_public_key(pr.synth_sk(i)) for i in range(n)]
