@@ -52,6 +52,12 @@ def _load():
                                             ctypes.POINTER(ctypes.c_int32)]
     lib.hbls_construct_commit_payload.argtypes = [ctypes.c_uint64, ctypes.c_char_p,
                                                   ctypes.c_uint64, ctypes.c_int, ctypes.c_char_p]
+    lib.hbls_mask_partials.argtypes = [ctypes.c_void_p, ctypes.c_char_p,
+                                       ctypes.c_size_t, ctypes.c_char_p]
+    lib.hbls_batch_agg_verify_partials.argtypes = [
+        ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t,
+        ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_size_t,
+        ctypes.POINTER(ctypes.c_int32)]
     return lib
 
 
@@ -208,6 +214,19 @@ class Committee:
         res = (ctypes.c_int32 * batch)()
         _check(_lib.hbls_batch_agg_verify(self._h, bitmaps, sigs, msgs, mlen,
                                           batch, res), "batch_agg_verify")
+        return list(res)
+
+    def mask_partials(self, bitmaps: bytes, batch: int) -> bytes:
+        out = ctypes.create_string_buffer(48 * batch)
+        _check(_lib.hbls_mask_partials(self._h, bitmaps, batch, out), "mask_partials")
+        return out.raw
+
+    def batch_agg_verify_partials(self, bitmaps: bytes, ext48s: bytes, n_ext: int,
+                                  sigs: bytes, msgs: bytes, mlen: int, batch: int):
+        res = (ctypes.c_int32 * batch)()
+        _check(_lib.hbls_batch_agg_verify_partials(
+            self._h, bitmaps, ext48s, n_ext, sigs, msgs, mlen, batch, res),
+            "batch_agg_verify_partials")
         return list(res)
 
     def batch_verify_votes(self, key_idx, sigs: bytes, msgs: bytes, mlen: int):

@@ -115,7 +115,7 @@ DEV void fp_neg(fp_t &r, const fp_t &x) {
 }
 DEV void fp_dbl(fp_t &r, const fp_t &x) { fp_add(r, x, x); }
 
-DEV void fp_mul(fp_t &r, const fp_t &x, const fp_t &y) {
+DEVN void fp_mul(fp_t &r, const fp_t &x, const fp_t &y) {
     uint64_t t[7];
 #pragma unroll
     for (int i = 0; i < 7; i++) t[i] = 0;
@@ -165,7 +165,7 @@ DEV void fp_to_mont(fp_t &r, const uint64_t raw[6]) {
     for (int i = 0; i < 6; i++) { t.l[i] = raw[i]; r2.l[i] = BLS_R2P[i]; }
     fp_mul(r, t, r2);
 }
-DEV void fp_from_mont(uint64_t raw[6], const fp_t &x) {
+DEVN void fp_from_mont(uint64_t raw[6], const fp_t &x) {
     fp_t one, t;
     fp_zero(one);
     one.l[0] = 1;
@@ -218,7 +218,7 @@ DEV void fp2_sub(fp2_t &r, const fp2_t &x, const fp2_t &y) { fp_sub(r.a, x.a, y.
 DEV void fp2_neg(fp2_t &r, const fp2_t &x) { fp_neg(r.a, x.a); fp_neg(r.b, x.b); }
 DEV void fp2_conj(fp2_t &r, const fp2_t &x) { r.a = x.a; fp_neg(r.b, x.b); }
 DEV void fp2_dbl(fp2_t &r, const fp2_t &x) { fp2_add(r, x, x); }
-DEV void fp2_mul(fp2_t &r, const fp2_t &x, const fp2_t &y) {
+DEVN void fp2_mul(fp2_t &r, const fp2_t &x, const fp2_t &y) {
     fp_t ac, bd, ab, cd, t;
     fp_mul(ac, x.a, y.a);
     fp_mul(bd, x.b, y.b);
@@ -230,7 +230,7 @@ DEV void fp2_mul(fp2_t &r, const fp2_t &x, const fp2_t &y) {
     fp_sub(r.a, ac, bd);
     r.b = t;
 }
-DEV void fp2_sqr(fp2_t &r, const fp2_t &x) {
+DEVN void fp2_sqr(fp2_t &r, const fp2_t &x) {
     fp_t s, d, m;
     fp_add(s, x.a, x.b);
     fp_sub(d, x.a, x.b);
@@ -497,7 +497,7 @@ DEVN void g2_psi(g2_t &r, const g2_t &p) {
 }
 
 /* ================================================================ serialization */
-DEV void fp_to_le48(uint8_t out[48], const fp_t &x) {
+DEVN void fp_to_le48(uint8_t out[48], const fp_t &x) {
     uint64_t raw[6];
     fp_from_mont(raw, x);
 #pragma unroll
@@ -506,7 +506,7 @@ DEV void fp_to_le48(uint8_t out[48], const fp_t &x) {
         for (int j = 0; j < 8; j++)
             out[i * 8 + j] = (uint8_t)(raw[i] >> (8 * j));
 }
-DEV bool fp_from_le48(fp_t &x, const uint8_t in[48]) {
+DEVN bool fp_from_le48(fp_t &x, const uint8_t in[48]) {
     uint64_t raw[6];
 #pragma unroll
     for (int i = 0; i < 6; i++) {
@@ -1272,6 +1272,24 @@ __global__ void k_g1_reduce(const g1_t *partials, int n, uint8_t *out48) {
     g1_serialize(out48, acc);
 }
 
+/* config-4 support: add n_ext serialized G1 partials (from other ranks'
+ * committee slices) into each item's aggregate before the pairing check.
+ * Partials are internal products (no subgroup check; identity = 48 zeros). */
+__global__ void k_add_partials(g1_t *aggs, const uint8_t *ext48s, int n_ext,
+                               int32_t *ok, int batch) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= batch) return;
+    g1_t acc = aggs[i];
+    for (int e = 0; e < n_ext; e++) {
+        const uint8_t *p = ext48s + ((size_t)e * batch + i) * 48;
+        g1_t q;
+        if (!g1_deserialize(q, p, false)) { ok[i] = 0; return; }
+        g1_add(acc, acc, q);
+    }
+    aggs[i] = acc;
+    ok[i] = 1;
+}
+
 __global__ void k_keccak(const uint8_t *msgs, int mlen, uint8_t *outs, int batch) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= batch) return;
@@ -1741,6 +1759,74 @@ extern "C" double hbls_mad_peak_ops(void) {
     (void)hipEventDestroy(e0); (void)hipEventDestroy(e1);
     double total_ops = (double)blocks * 256.0 * (double)iters * 8.0 * 4.0;
     return total_ops / (ms * 1e-3);
+}
+
+/* config-4: local masked partial sums, serialized (one 48B point per item) */
+extern "C" int hbls_mask_partials(const hbls_committee_t *c, const uint8_t *bitmaps,
+                                  size_t batch, uint8_t *out48s) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    size_t bm = (c->n + 7) / 8;
+    DevBuf dbm(batch * bm), dagg(batch * sizeof(g1_t)), dser(batch * 48);
+    if (dbm.err || dagg.err || dser.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(dbm.p, bitmaps, batch * bm, hipMemcpyHostToDevice));
+    Timer tm;
+    int nb = (int)((batch + 63) / 64);
+    hipLaunchKernelGGL(k_mask_aggregate, dim3((int)batch), dim3(MASK_BLOCK), 0, 0,
+                       c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
+                       dagg.as<g1_t>(), (int)batch);
+    hipLaunchKernelGGL(k_g1_serialize, dim3(nb), dim3(64), 0, 0,
+                       dagg.as<g1_t>(), dser.as<uint8_t>(), (int)batch);
+    tm.stop_and_store();
+    HIP_OK(hipGetLastError());
+    HIP_OK(hipMemcpy(out48s, dser.p, batch * 48, hipMemcpyDeviceToHost));
+    return HBLS_OK;
+}
+
+/* config-4: aggregate-verify where this rank holds ONE SLICE of the
+ * committee; ext48s carries the other ranks' per-item partial sums
+ * (n_ext blocks of batch x 48 B, layout [ext][item]). */
+extern "C" int hbls_batch_agg_verify_partials(
+        const hbls_committee_t *c, const uint8_t *bitmaps,
+        const uint8_t *ext48s, size_t n_ext,
+        const uint8_t *sigs96, const uint8_t *msgs,
+        size_t msg_len, size_t batch, int32_t *results) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    size_t bm = (c->n + 7) / 8;
+    DevBuf dbm(batch * bm), dext(batch * 48 * (n_ext ? n_ext : 1));
+    DevBuf dsig(batch * 96), dmsg(batch * msg_len);
+    DevBuf dagg(batch * sizeof(g1_t)), dhm(batch * sizeof(g2_t));
+    DevBuf dsaff(batch * sizeof(g2aff_t)), dsflags(batch * 4), dhok(batch * 4);
+    DevBuf dres(batch * 4), dpok(batch * 4);
+    if (dbm.err || dext.err || dsig.err || dmsg.err || dagg.err || dhm.err ||
+        dsaff.err || dsflags.err || dhok.err || dres.err || dpok.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(dbm.p, bitmaps, batch * bm, hipMemcpyHostToDevice));
+    if (n_ext)
+        HIP_OK(hipMemcpy(dext.p, ext48s, batch * 48 * n_ext, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(dsig.p, sigs96, batch * 96, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(dmsg.p, msgs, batch * msg_len, hipMemcpyHostToDevice));
+    Timer tm;
+    int nb = (int)((batch + 63) / 64);
+    hipLaunchKernelGGL(k_mask_aggregate, dim3((int)batch), dim3(MASK_BLOCK), 0, 0,
+                       c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
+                       dagg.as<g1_t>(), (int)batch);
+    if (n_ext)
+        hipLaunchKernelGGL(k_add_partials, dim3(nb), dim3(64), 0, 0,
+                           dagg.as<g1_t>(), dext.as<uint8_t>(), (int)n_ext,
+                           dpok.as<int32_t>(), (int)batch);
+    hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
+                       dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+                       (int)batch, g_fast_cofactor);
+    hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
+                       dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0,
+                       dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+                       dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
+    tm.stop_and_store();
+    HIP_OK(hipGetLastError());
+    HIP_OK(hipMemcpy(results, dres.p, batch * 4, hipMemcpyDeviceToHost));
+    return HBLS_OK;
 }
 
 /* ConstructCommitPayload (consensus/signature/signature.go:12-24):

@@ -129,6 +129,19 @@ int hbls_batch_sign(const uint8_t *sks32, const uint8_t *msgs, size_t msg_len,
 /* batch sk->pk */
 int hbls_batch_pk_from_sk(const uint8_t *sks32, size_t batch, uint8_t *pks48);
 
+/* ---- config-4 (sharded 65536-key committee across GPUs) ----
+ * Each rank holds one index-range slice of the committee as its table.
+ * hbls_mask_partials computes this rank's per-item masked partial sums
+ * (serialized, batch x 48 B) for the RCCL all-gather; the receiving rank
+ * folds the other ranks' partials into its own slice's sums and finishes
+ * the pairing check. */
+int hbls_mask_partials(const hbls_committee_t *c, const uint8_t *bitmaps,
+                       size_t batch, uint8_t *out48s);
+int hbls_batch_agg_verify_partials(const hbls_committee_t *c, const uint8_t *bitmaps,
+                                   const uint8_t *ext48s, size_t n_ext,
+                                   const uint8_t *sigs96, const uint8_t *msgs,
+                                   size_t msg_len, size_t batch, int32_t *results);
+
 /* G1 MSM: out = sum_i scalar_i * P_i (scalars 32B LE each; general building
  * block behind mask aggregation; Pippenger on device) */
 int hbls_msm_g1(const uint8_t *points48, const uint8_t *scalars32, size_t n,

@@ -250,6 +250,44 @@ def test_keccak_batch_parity(core, capi):
         assert got[32 * j:32 * j + 32] == capi.keccak256(msgs[mlen * j:mlen * (j + 1)])
 
 
+def test_config4_sharded_partials(core, capi):
+    """config-4 shape on one GPU: committee split into two slices; slice A's
+    partial sums travel serialized (the RCCL payload) and are folded into
+    slice B's verify — result must equal the full-committee verify."""
+    n = 32
+    sks = b"".join(sk_bytes(i) for i in range(n))
+    pks = core.batch_pk_from_sk(sks, n)
+    half = n // 2
+    cA = core.Committee(pks[:48 * half], half)
+    cB = core.Committee(pks[48 * half:], half)
+    full = core.Committee(pks, n)
+    batch = 3
+    bmlen = (n + 7) // 8
+    bmsA, bmsB, bms, sigs, msgs = b"", b"", b"", b"", b""
+    for j in range(batch):
+        msg = pr.construct_commit_payload(j, pr.synth_msg(j), j)
+        signers = [i for i in range(n) if (i * 5 + j) % 4 != 0]
+        bm = bytearray(bmlen)
+        for i in signers:
+            bm[i >> 3] |= 1 << (i & 7)
+        sk_sum = sum(pr.synth_sk(i) for i in signers) % pr.R
+        sig = capi.sign_hash(pr.fr_serialize(sk_sum), msg)
+        bms += bytes(bm)
+        bmsA += bytes(bm[:half // 8])
+        bmsB += bytes(bm[half // 8:])
+        sigs += sig
+        msgs += msg
+    partialsA = cA.mask_partials(bmsA, batch)
+    res = cB.batch_agg_verify_partials(bmsB, partialsA, 1, sigs, msgs, 48, batch)
+    assert res == [1, 1, 1]
+    assert full.batch_agg_verify(bms, sigs, msgs, 48, batch) == [1, 1, 1]
+    # corrupt one partial -> that item must reject
+    bad = bytearray(partialsA)
+    bad[0:48] = capi.pk_from_sk(sk_bytes(31))
+    res2 = cB.batch_agg_verify_partials(bmsB, bytes(bad), 1, sigs, msgs, 48, batch)
+    assert res2[0] == 0 and res2[1:] == [1, 1]
+
+
 # --------------------------------------------------- config-2 size (4096) identities
 @pytest.fixture(scope="module")
 def committee4096(core):
