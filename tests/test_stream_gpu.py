@@ -48,8 +48,9 @@ def test_stream_round(oracle_lib):
         for i in chunk:
             s = sigs_all[96 * i:96 * (i + 1)]
             if i == bad_at:
-                # substitute wrong-message signature
-                s = oracle_lib.sign_hash(sks[i], payload[:-1] + b"\x00")
+                # substitute wrong-message signature (flip a hash byte —
+                # the payload TAIL is LE64(viewID) and may already be zero)
+                s = oracle_lib.sign_hash(sks[i], bytes([payload[0] ^ 0xFF]) + payload[1:])
             sigs += s
         blobs = b"".join((pr.synth_msg(i) * 20)[:blob_len] for i in chunk)
         sv.process_batch(chunk, sigs, blobs, blob_len)
