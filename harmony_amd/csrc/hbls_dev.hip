@@ -789,7 +789,18 @@ DEVN void fp12_mul(fp12_t &r, const fp12_t &x, const fp12_t &y) {
     fp6_add(r.c0, t0, vt1);
     r.c1 = tt;
 }
-DEV void fp12_sqr(fp12_t &r, const fp12_t &x) { fp12_mul(r, x, x); }
+DEVN void fp12_sqr(fp12_t &r, const fp12_t &x) {
+    fp6_t ab, apb, avb, t0, vab;
+    fp6_mul(ab, x.c0, x.c1);
+    fp6_add(apb, x.c0, x.c1);
+    fp6_mul_v(avb, x.c1);
+    fp6_add(avb, x.c0, avb);
+    fp6_mul(t0, apb, avb);
+    fp6_sub(t0, t0, ab);
+    fp6_mul_v(vab, ab);
+    fp6_sub(r.c0, t0, vab);
+    fp6_add(r.c1, ab, ab);
+}
 DEV void fp12_conj(fp12_t &r, const fp12_t &x) { r.c0 = x.c0; fp6_neg(r.c1, x.c1); }
 DEVN void fp12_inv(fp12_t &r, const fp12_t &x) {
     fp6_t t, t1;
@@ -859,12 +870,33 @@ DEVN void fp12_frob2(fp12_t &r, const fp12_t &x) {
     r = tmp;
 }
 DEVN void fp12_mul_line(fp12_t &f, const fp2_t &c0, const fp2_t &c3, const fp2_t &c5) {
-    fp12_t l;
-    fp2_zero(l.c0.c1); fp2_zero(l.c0.c2); fp2_zero(l.c1.c0);
-    l.c0.c0 = c0;
-    l.c1.c1 = c3;
-    l.c1.c2 = c5;
-    fp12_mul(f, f, l);
+    fp6_t t0, t1, l01, tt, vt1, fs, l;
+    fp2_mul(t0.c0, f.c0.c0, c0);
+    fp2_mul(t0.c1, f.c0.c1, c0);
+    fp2_mul(t0.c2, f.c0.c2, c0);
+    {
+        const fp2_t &a0 = f.c1.c0, &a1 = f.c1.c1, &a2 = f.c1.c2;
+        fp2_t p1, p2, q;
+        fp2_mul(p1, a1, c5);
+        fp2_mul(p2, a2, c3);
+        fp2_add(q, p1, p2);
+        fp2_mul_xi(t1.c0, q);
+        fp2_mul(p1, a0, c3);
+        fp2_mul(p2, a2, c5);
+        fp2_mul_xi(p2, p2);
+        fp2_add(t1.c1, p1, p2);
+        fp2_mul(p1, a0, c5);
+        fp2_mul(p2, a1, c3);
+        fp2_add(t1.c2, p1, p2);
+    }
+    fp6_add(fs, f.c0, f.c1);
+    l.c0 = c0; l.c1 = c3; l.c2 = c5;
+    fp6_mul(l01, fs, l);
+    fp6_sub(tt, l01, t0);
+    fp6_sub(tt, tt, t1);
+    fp6_mul_v(vt1, t1);
+    fp6_add(f.c0, t0, vt1);
+    f.c1 = tt;
 }
 /* f_{|z|,Q}(P); accumulate into f (caller inits f=1 or continues a product) */
 DEVN void miller_loop_acc(fp12_t &f, const g2aff_t &Q, const g1aff_t &Pa) {
@@ -924,10 +956,40 @@ DEVN void miller_loop_acc(fp12_t &f, const g2aff_t &Q, const g1aff_t &Pa) {
         }
     }
 }
+/* Granger-Scott cyclotomic squaring (cyclotomic-subgroup elements only;
+ * validated against fp12_sqr by the oracle self-test + GPU parity) */
+DEV void fp4_sqr_gs(fp2_t &c, fp2_t &d, const fp2_t &a, const fp2_t &b) {
+    fp2_t a2, b2, t;
+    fp2_sqr(a2, a);
+    fp2_sqr(b2, b);
+    fp2_mul_xi(t, b2);
+    fp2_add(c, a2, t);
+    fp2_add(t, a, b);
+    fp2_sqr(t, t);
+    fp2_sub(t, t, a2);
+    fp2_sub(d, t, b2);
+}
+DEVN void fp12_cyc_sqr(fp12_t &r, const fp12_t &x) {
+    const fp2_t &a0 = x.c0.c0, &a1 = x.c1.c0, &a2 = x.c0.c1,
+                &a3 = x.c1.c1, &a4 = x.c0.c2, &a5 = x.c1.c2;
+    fp2_t t00, t03, t01, t04, t02, t05, tmp, x05;
+    fp4_sqr_gs(t00, t03, a0, a3);
+    fp4_sqr_gs(t01, t04, a1, a4);
+    fp4_sqr_gs(t02, t05, a2, a5);
+    fp12_t out;
+    fp2_sub(tmp, t00, a0); fp2_dbl(tmp, tmp); fp2_add(out.c0.c0, tmp, t00);
+    fp2_sub(tmp, t01, a2); fp2_dbl(tmp, tmp); fp2_add(out.c0.c1, tmp, t01);
+    fp2_sub(tmp, t02, a4); fp2_dbl(tmp, tmp); fp2_add(out.c0.c2, tmp, t02);
+    fp2_mul_xi(x05, t05);
+    fp2_add(tmp, x05, a1); fp2_dbl(tmp, tmp); fp2_add(out.c1.c0, tmp, x05);
+    fp2_add(tmp, t03, a3); fp2_dbl(tmp, tmp); fp2_add(out.c1.c1, tmp, t03);
+    fp2_add(tmp, t04, a5); fp2_dbl(tmp, tmp); fp2_add(out.c1.c2, tmp, t04);
+    r = out;
+}
 DEVN void fp12_pow_u(fp12_t &r, const fp12_t &x) {
     fp12_t acc = x;
     for (int bit = 62; bit >= 0; bit--) {
-        fp12_sqr(acc, acc);
+        fp12_cyc_sqr(acc, acc);
         if ((BLS_U >> bit) & 1) fp12_mul(acc, acc, x);
     }
     r = acc;

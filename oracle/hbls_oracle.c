@@ -473,13 +473,17 @@ static void g2_psi_aff(g2aff_t *r, const g2aff_t *p) {
     fp2_conj(&t, &p->x); fp2_mul(&r->x, &t, &cx);
     fp2_conj(&t, &p->y); fp2_mul(&r->y, &t, &cy);
 }
-static void g2_psi(g2_t *r, const g2_t *p) {
+static void g2_psi_jac(g2_t *r, const g2_t *p) {
+    /* x=X/Z^2, y=Y/Z^3 -> (cx*conj(X), cy*conj(Y), conj(Z)): no inversion */
     if (g2_is_inf(p)) { *r = *p; return; }
-    g2aff_t a, pa;
-    g2_to_affine(&a, p);
-    g2_psi_aff(&pa, &a);
-    g2_from_affine(r, &pa);
+    fp2_t cx, cy, t;
+    memcpy(cx.a.l, BLS_PSI_CX_A, 48); memcpy(cx.b.l, BLS_PSI_CX_B, 48);
+    memcpy(cy.a.l, BLS_PSI_CY_A, 48); memcpy(cy.b.l, BLS_PSI_CY_B, 48);
+    fp2_conj(&t, &p->x); fp2_mul(&r->x, &t, &cx);
+    fp2_conj(&t, &p->y); fp2_mul(&r->y, &t, &cy);
+    fp2_conj(&r->z, &p->z);
 }
+static void g2_psi(g2_t *r, const g2_t *p) { g2_psi_jac(r, p); }
 
 /* ================================================================== serialization */
 static void fp_to_le48(uint8_t out[48], const fp_t *x) {
@@ -530,6 +534,19 @@ static int g2_in_subgroup(const g2_t *p) {
     g2_t t;
     g2_mul(&t, p, BLS_R, 4);
     return g2_is_inf(&t);
+}
+static void g2_psi_jac(g2_t *r, const g2_t *p);
+static int g2_eq(const g2_t *p, const g2_t *q);
+/* Scott's criterion: Q in G2 iff psi(Q) == [z]Q (z<0) — same accept set as
+ * the [r]Q test (equivalence property-tested in tests/test_oracle.py) */
+static int g2_in_subgroup_fast(const g2_t *p) {
+    if (g2_is_inf(p)) return 1;
+    g2_t lhs, rhs;
+    g2_psi_jac(&lhs, p);
+    uint64_t u = BLS_U;
+    g2_mul(&rhs, p, &u, 1);
+    g2_neg(&rhs, &rhs);
+    return g2_eq(&lhs, &rhs);
 }
 
 static int g1_on_curve(const g1aff_t *a) {
@@ -776,7 +793,20 @@ static void fp12_mul(fp12_t *r, const fp12_t *x, const fp12_t *y) {
     fp6_add(&r->c0, &t0, &vt1);
     r->c1 = tt;
 }
-static void fp12_sqr(fp12_t *r, const fp12_t *x) { fp12_mul(r, x, x); }
+static void fp12_sqr(fp12_t *r, const fp12_t *x) {
+    /* (a + b w)^2 = (a^2 + v b^2) + 2ab w, via
+       a^2 + v b^2 = (a + b)(a + v b) - ab - v ab : 2 fp6 muls total */
+    fp6_t ab, apb, avb, t0, vab;
+    fp6_mul(&ab, &x->c0, &x->c1);
+    fp6_add(&apb, &x->c0, &x->c1);
+    fp6_mul_v(&avb, &x->c1);
+    fp6_add(&avb, &x->c0, &avb);
+    fp6_mul(&t0, &apb, &avb);
+    fp6_sub(&t0, &t0, &ab);
+    fp6_mul_v(&vab, &ab);
+    fp6_sub(&r->c0, &t0, &vab);
+    fp6_add(&r->c1, &ab, &ab);
+}
 static void fp12_conj(fp12_t *r, const fp12_t *x) { r->c0 = x->c0; fp6_neg(&r->c1, &x->c1); }
 static void fp12_inv(fp12_t *r, const fp12_t *x) {
     fp6_t t, t1;
@@ -834,12 +864,43 @@ static void fp12_frob2(fp12_t *r, const fp12_t *x) {
  *   ADD: c0 = xi*yP*ZH,    c3 = r*x2 - y2*Z*H, c5 = -r*xP
  * (derivation in DESIGN.md; validated against pyref's embedded-point loop) */
 static void fp12_mul_line(fp12_t *f, const fp2_t *c0, const fp2_t *c3, const fp2_t *c5) {
-    fp12_t l;
-    memset(&l, 0, sizeof(l));
-    l.c0.c0 = *c0;    /* w^0 */
-    l.c1.c1 = *c3;    /* w^3 = w*v */
-    l.c1.c2 = *c5;    /* w^5 = w*v^2 */
-    fp12_mul(f, f, &l);
+    /* sparse multiply by l = c0 + c3 w^3 + c5 w^5 (l0 = (c0,0,0), l1 = (0,c3,c5)) */
+    fp6_t t0, t1, l01, tt, vt1;
+    /* t0 = f0 * l0 : scale by c0 */
+    fp2_mul(&t0.c0, &f->c0.c0, c0);
+    fp2_mul(&t0.c1, &f->c0.c1, c0);
+    fp2_mul(&t0.c2, &f->c0.c2, c0);
+    /* t1 = f1 * l1 with l1 = (0, c3, c5):
+       r0 = xi*(a1*c5 + a2*c3); r1 = xi*a2*c5; r2 = a0*c3 ... careful:
+       fp6 mul (a0,a1,a2)*(0,b1,b2) with v^3=xi:
+         r0 = xi*(a1*b2 + a2*b1)
+         r1 = a0*b1 + xi*(a2*b2)
+         r2 = a0*b2 + a1*b1 */
+    {
+        const fp2_t *a0 = &f->c1.c0, *a1 = &f->c1.c1, *a2 = &f->c1.c2;
+        fp2_t p1, p2, q;
+        fp2_mul(&p1, a1, c5);
+        fp2_mul(&p2, a2, c3);
+        fp2_add(&q, &p1, &p2);
+        fp2_mul_xi(&t1.c0, &q);
+        fp2_mul(&p1, a0, c3);
+        fp2_mul(&p2, a2, c5);
+        fp2_mul_xi(&p2, &p2);
+        fp2_add(&t1.c1, &p1, &p2);
+        fp2_mul(&p1, a0, c5);
+        fp2_mul(&p2, a1, c3);
+        fp2_add(&t1.c2, &p1, &p2);
+    }
+    /* r1 = (f0+f1)*(l0+l1) - t0 - t1 ; l0+l1 = (c0, c3, c5) */
+    fp6_t fs, l;
+    fp6_add(&fs, &f->c0, &f->c1);
+    l.c0 = *c0; l.c1 = *c3; l.c2 = *c5;
+    fp6_mul(&l01, &fs, &l);
+    fp6_sub(&tt, &l01, &t0);
+    fp6_sub(&tt, &tt, &t1);
+    fp6_mul_v(&vt1, &t1);
+    fp6_add(&f->c0, &t0, &vt1);
+    f->c1 = tt;
 }
 static void miller_loop(fp12_t *f, const g2aff_t *Q, const g1aff_t *Pa) {
     fp12_one(f);
@@ -917,11 +978,68 @@ static void miller_loop(fp12_t *f, const g2aff_t *Q, const g1aff_t *Pa) {
     }
 }
 
-/* exp by |z| using plain square-multiply (cyclotomic-safe: conj used by caller) */
+/* Granger-Scott cyclotomic squaring (valid for elements of the cyclotomic
+ * subgroup, i.e. after the easy part of the final exponentiation).
+ * Coefficients by w-power: a_i at [c0.c0, c1.c0, c0.c1, c1.c1, c0.c2, c1.c2].
+ * Working on the Fp4 towers (a0,a3), (a1,a4), (a2,a5):
+ *   fp4_sqr((a,b)) = (a^2 + xi*b^2... ) per GS2010. */
+static void fp4_sqr(fp2_t *c, fp2_t *d, const fp2_t *a, const fp2_t *b) {
+    /* (c + d*t) = (a + b*t)^2 in Fp4 = Fp2[t]/(t^2 - xi):
+       c = a^2 + xi*b^2, d = 2ab */
+    fp2_t a2, b2, t;
+    fp2_sqr(&a2, a);
+    fp2_sqr(&b2, b);
+    fp2_mul_xi(&t, &b2);
+    fp2_add(c, &a2, &t);
+    fp2_add(&t, a, b);
+    fp2_sqr(&t, &t);
+    fp2_sub(&t, &t, &a2);
+    fp2_sub(d, &t, &b2);
+}
+static void fp12_cyc_sqr(fp12_t *r, const fp12_t *x) {
+    const fp2_t *a0 = &x->c0.c0, *a1 = &x->c1.c0, *a2 = &x->c0.c1,
+                *a3 = &x->c1.c1, *a4 = &x->c0.c2, *a5 = &x->c1.c2;
+    fp2_t t00, t03, t01, t04, t02, t05, t;
+    fp4_sqr(&t00, &t03, a0, a3);   /* (a0 + a3 t) */
+    fp4_sqr(&t01, &t04, a1, a4);   /* (a1 + a4 t) */
+    fp4_sqr(&t02, &t05, a2, a5);   /* (a2 + a5 t) */
+    /* r0 = 3 t00 - 2 a0 ; r3 = 3 t03 + 2 a3
+       r2 = 3 t01 - 2 a2 ; r5 = 3 t04 + 2 a5   (shifted by the w-multiplication)
+       r4 = 3 t02 - 2 a4 ; r1 = 3 xi*t05 + 2 a1 */
+    fp12_t out;
+#define GS(dst, tv, av, neg_) do { \
+    fp2_t s; \
+    fp2_sub(&s, (tv), (av)); \
+    if (neg_) fp2_neg(&s, &s); \
+    fp2_dbl(&s, &s); \
+    fp2_add(&s, &s, (tv)); \
+    dst = s; } while (0)
+    /* dst = 2*(t - a) + t  (for minus case)  /  2*(t + a) + t (plus case) */
+    fp2_t tmp;
+    /* r0 = 2(t00 - a0) + t00 */
+    fp2_sub(&tmp, &t00, a0); fp2_dbl(&tmp, &tmp); fp2_add(&out.c0.c0, &tmp, &t00);
+    /* r2 (coeff a2, at w^2) = 2(t01 - a2) + t01 */
+    fp2_sub(&tmp, &t01, a2); fp2_dbl(&tmp, &tmp); fp2_add(&out.c0.c1, &tmp, &t01);
+    /* r4 (a4, w^4) = 2(t02 - a4) + t02 */
+    fp2_sub(&tmp, &t02, a4); fp2_dbl(&tmp, &tmp); fp2_add(&out.c0.c2, &tmp, &t02);
+    /* r1 (a1, w^1) = 2(xi*t05 + a1) + xi*t05 */
+    fp2_t x05;
+    fp2_mul_xi(&x05, &t05);
+    fp2_add(&tmp, &x05, a1); fp2_dbl(&tmp, &tmp); fp2_add(&out.c1.c0, &tmp, &x05);
+    /* r3 (a3, w^3) = 2(t03 + a3) + t03 */
+    fp2_add(&tmp, &t03, a3); fp2_dbl(&tmp, &tmp); fp2_add(&out.c1.c1, &tmp, &t03);
+    /* r5 (a5, w^5) = 2(t04 + a5) + t04 */
+    fp2_add(&tmp, &t04, a5); fp2_dbl(&tmp, &tmp); fp2_add(&out.c1.c2, &tmp, &t04);
+#undef GS
+    (void)t;
+    *r = out;
+}
+/* exp by |z|; cyclotomic squarings (callers apply this only after the easy
+ * part, where x is in the cyclotomic subgroup). */
 static void fp12_pow_u(fp12_t *r, const fp12_t *x) {
     fp12_t acc = *x;
     for (int bit = 62; bit >= 0; bit--) {
-        fp12_sqr(&acc, &acc);
+        fp12_cyc_sqr(&acc, &acc);
         if ((BLS_U >> bit) & 1) fp12_mul(&acc, &acc, x);
     }
     *r = acc;
@@ -975,6 +1093,43 @@ int oracle_pairing_check2(const g2aff_t *Q1, const g1aff_t *P1,
     fp12_conj(&f, &f);
     final_exp(&f, &f);
     return fp12_is_one(&f);
+}
+
+/* test support: both G2 membership methods on a compressed candidate */
+int oracle_g2_subgroup_methods(const uint8_t in96[96], int32_t out2[2]) {
+    g2_t p;
+    if (!oracle_g2_deserialize(&p, in96, 0)) { out2[0] = out2[1] = -1; return 0; }
+    out2[0] = g2_in_subgroup(&p);
+    out2[1] = g2_in_subgroup_fast(&p);
+    return 1;
+}
+
+/* self-test: cyclotomic squaring must equal the full squaring on elements of
+ * the cyclotomic subgroup (exercised by tests/test_oracle.py) */
+int oracle_test_cyc_sqr(void) {
+    g1_t base;
+    g1aff_t ba;
+    g2aff_t qa;
+    fp_base_point(&base);
+    g1_to_affine(&ba, &base);
+    {
+        g2_t q;
+        fp2_t xa, ya;
+        memcpy(xa.a.l, BLS_G2_XA, 48); memcpy(xa.b.l, BLS_G2_XB, 48);
+        memcpy(ya.a.l, BLS_G2_YA, 48); memcpy(ya.b.l, BLS_G2_YB, 48);
+        qa.x = xa; qa.y = ya;
+        (void)q;
+    }
+    fp12_t f, easy, t, inv, a, b;
+    miller_loop(&f, &qa, &ba);
+    fp12_conj(&t, &f);
+    fp12_inv(&inv, &f);
+    fp12_mul(&easy, &t, &inv);
+    fp12_frob2(&t, &easy);
+    fp12_mul(&easy, &t, &easy);
+    fp12_cyc_sqr(&a, &easy);
+    fp12_sqr(&b, &easy);
+    return fp12_eq(&a, &b);
 }
 
 /* ================================================================== Keccak-256 */
