@@ -149,9 +149,15 @@ class Mask:
         self.Publics = list(publics)            # list[PublicKeyWrapper]
         self.PublicsIndex = {w.Bytes: i for i, w in enumerate(self.Publics)}
         self.Bitmap = bytearray(self.length())
-        self._committee = core.Committee(
-            b"".join(w.Bytes for w in self.Publics), len(self.Publics))
+        self._committee_cache = None            # lazy: bit bookkeeping needs no GPU
         self._agg_cache = None
+
+    @property
+    def _committee(self):
+        if self._committee_cache is None:
+            self._committee_cache = core.Committee(
+                b"".join(w.Bytes for w in self.Publics), len(self.Publics))
+        return self._committee_cache
 
     def length(self) -> int:
         return (len(self.Publics) + 7) >> 3

@@ -71,8 +71,14 @@ class Decider:
         if stakes is not None:
             total = sum(Fraction(s) for s in stakes)
             self.stakes = [Fraction(s) / total for s in stakes]
-        self._committee = core.Committee(
-            b"".join(w.Bytes for w in self.members), len(self.members))
+        self._committee_cache = None   # built lazily: host-only logic needs no GPU
+
+    @property
+    def _committee(self):
+        if self._committee_cache is None:
+            self._committee_cache = core.Committee(
+                b"".join(w.Bytes for w in self.members), len(self.members))
+        return self._committee_cache
 
     # -- policy (one-node-one-vote.go:42-58 / one-node-staked-vote.go:175-188)
     def two_thirds_count(self) -> int:
