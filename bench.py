@@ -109,11 +109,14 @@ def run_config4(args, rank, world, dist):
     my_msgs = msgs[MSG_LEN * i0:MSG_LEN * (i0 + share)]
 
     import torch
+    backend = os.environ.get("HBLS_DIST_BACKEND", "nccl")
 
     def one_step(timed_check=True):
         partials = slice_table.mask_partials(slice_bms, batch)   # batch x 48
         if world > 1:
-            t = torch.frombuffer(bytearray(partials), dtype=torch.uint8).cuda()
+            t = torch.frombuffer(bytearray(partials), dtype=torch.uint8)
+            if backend == "nccl":
+                t = t.cuda()
             outs = [torch.empty_like(t) for _ in range(world)]
             dist.all_gather(outs, t)
             ext = b"".join(o.cpu().numpy().tobytes()
@@ -135,7 +138,8 @@ def run_config4(args, rank, world, dist):
     def barrier_sync():
         if dist is not None:
             dist.barrier()
-            torch.cuda.synchronize()
+            if backend == "nccl":
+                torch.cuda.synchronize()
     for _ in range(args.warmup):
         one_step(False)
     barrier_sync()
@@ -146,7 +150,8 @@ def run_config4(args, rank, world, dist):
     t1 = time.perf_counter()
     elapsed = t1 - t0
     if dist is not None:
-        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if backend == "nccl" else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
     if rank != 0:
@@ -173,7 +178,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=int(os.environ.get("HBLS_BENCH_BATCH", "16384")))
+    ap.add_argument("--batch", type=int, default=int(os.environ.get("HBLS_BENCH_BATCH", "49152")))
     ap.add_argument("--mode", choices=["config2", "config4"], default="config2",
                     help="config2: per-rank replica committees (default, weak scaling); "
                          "config4: one 65536-key committee sharded across ranks with "
@@ -185,19 +190,22 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
 
+    backend = os.environ.get("HBLS_DIST_BACKEND", "nccl")
     dist = None
     if world > 1:
         import torch
         import torch.distributed as tdist
-        tdist.init_process_group(backend="nccl")
-        torch.cuda.set_device(local_rank)
+        tdist.init_process_group(backend=backend)
+        if backend == "nccl":
+            torch.cuda.set_device(local_rank)
         dist = tdist
 
     from harmony_amd import core
     if core.device_count() == 0:
         print(json.dumps({"error": "no AMD GPU — bench requires MI355X"}))
         sys.exit(1)
-    core.init(local_rank if world > 1 else -1)
+    # modulo: lets a world>1 gloo dry-run share one GPU on a 1-GPU box
+    core.init(local_rank % core.device_count() if world > 1 else -1)
 
     from oracle import capi, pyref as pr
 
@@ -250,7 +258,8 @@ def main():
     # max over ranks
     if dist is not None:
         import torch
-        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dev = "cuda" if backend == "nccl" else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
