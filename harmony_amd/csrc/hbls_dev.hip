@@ -1772,6 +1772,126 @@ extern "C" int hbls_msm_g1(const uint8_t *points48, const uint8_t *scalars32, si
     return HBLS_OK;
 }
 
+/* ---- fp_mul representation A/B microbenchmark ----
+ * Variant 0: the shipping 6x64-limb CIOS (u128 accumulators).
+ * Variant 1: 12x32-limb CIOS — every product is a 32x32+64 mad, the native
+ *   v_mad_u64_u32 shape; more mads (288 vs 72 wider ones) but no carry
+ *   juggling beyond the accumulator flow. */
+DEV void fp_mul32(uint32_t r[12], const uint32_t a[12], const uint32_t b[12],
+                  const uint32_t p32[12], uint32_t pinv32) {
+    uint32_t t[13];
+#pragma unroll
+    for (int i = 0; i < 13; i++) t[i] = 0;
+    uint32_t t13 = 0;
+#pragma unroll
+    for (int i = 0; i < 12; i++) {
+        uint64_t acc = 0;
+        uint32_t ai = a[i];
+#pragma unroll
+        for (int j = 0; j < 12; j++) {
+            acc = (uint64_t)ai * b[j] + t[j] + (uint32_t)(acc >> 32);
+            t[j] = (uint32_t)acc;
+        }
+        acc = (uint64_t)t[12] + (uint32_t)(acc >> 32);
+        t[12] = (uint32_t)acc;
+        t13 = (uint32_t)(acc >> 32);
+        uint32_t m = t[0] * pinv32;
+        acc = (uint64_t)m * p32[0] + t[0];
+#pragma unroll
+        for (int j = 1; j < 12; j++) {
+            acc = (uint64_t)m * p32[j] + t[j] + (uint32_t)(acc >> 32);
+            t[j - 1] = (uint32_t)acc;
+        }
+        acc = (uint64_t)t[12] + (uint32_t)(acc >> 32);
+        t[11] = (uint32_t)acc;
+        t[12] = t13 + (uint32_t)(acc >> 32);
+    }
+    /* conditional subtract p */
+    bool ge = t[12] != 0;
+    if (!ge) {
+        ge = true;
+        for (int i = 11; i >= 0; i--) {
+            if (t[i] > p32[i]) { ge = true; break; }
+            if (t[i] < p32[i]) { ge = false; break; }
+        }
+    }
+    if (ge) {
+        uint64_t bw = 0;
+#pragma unroll
+        for (int i = 0; i < 12; i++) {
+            uint64_t d = (uint64_t)t[i] - p32[i] - (uint32_t)bw;
+            r[i] = (uint32_t)d;
+            bw = (d >> 32) & 1;
+        }
+    } else {
+#pragma unroll
+        for (int i = 0; i < 12; i++) r[i] = t[i];
+    }
+}
+__global__ void __launch_bounds__(256) k_fpmul_bench(uint64_t *sink, int iters, int variant) {
+    /* independent 4-chain per thread to expose ILP, like the real kernels */
+    fp_t a, b;
+#pragma unroll
+    for (int i = 0; i < 6; i++) {
+        a.l[i] = BLS_ONE_P[i] ^ (blockIdx.x * 256 + threadIdx.x);
+        b.l[i] = BLS_R2P[i] ^ (i * 1234567u);
+    }
+    a.l[5] &= 0x0fffffffffffffffULL;
+    b.l[5] &= 0x0fffffffffffffffULL;
+    if (variant == 0) {
+        fp_t x0 = a, x1 = b, x2 = a, x3 = b;
+        for (int it = 0; it < iters; it++) {
+            fp_mul(x0, x0, a);
+            fp_mul(x1, x1, b);
+            fp_mul(x2, x2, a);
+            fp_mul(x3, x3, b);
+        }
+        if (x0.l[0] == 0xdeadbeef) sink[threadIdx.x] = x0.l[0] + x1.l[1] + x2.l[2] + x3.l[3];
+    } else {
+        uint32_t p32[12], a32[12], b32[12], x0[12], x1[12], x2[12], x3[12];
+#pragma unroll
+        for (int i = 0; i < 6; i++) {
+            p32[2 * i] = (uint32_t)BLS_P[i];
+            p32[2 * i + 1] = (uint32_t)(BLS_P[i] >> 32);
+            a32[2 * i] = (uint32_t)a.l[i];
+            a32[2 * i + 1] = (uint32_t)(a.l[i] >> 32);
+            b32[2 * i] = (uint32_t)b.l[i];
+            b32[2 * i + 1] = (uint32_t)(b.l[i] >> 32);
+        }
+        uint32_t pinv32 = (uint32_t)BLS_P_INV;   /* -p^-1 mod 2^32 = low word */
+#pragma unroll
+        for (int i = 0; i < 12; i++) { x0[i] = a32[i]; x1[i] = b32[i]; x2[i] = a32[i]; x3[i] = b32[i]; }
+        for (int it = 0; it < iters; it++) {
+            fp_mul32(x0, x0, a32, p32, pinv32);
+            fp_mul32(x1, x1, b32, p32, pinv32);
+            fp_mul32(x2, x2, a32, p32, pinv32);
+            fp_mul32(x3, x3, b32, p32, pinv32);
+        }
+        if (x0[0] == 0xdeadbeef) sink[threadIdx.x] = x0[0] + x1[1] + x2[2] + x3[3];
+    }
+}
+/* measured fp_mul throughput (muls/s) for the given variant */
+extern "C" double hbls_fpmul_bench_ops(int variant) {
+    if (require_gpu() != HBLS_OK) return 0.0;
+    DevBuf sink(256 * 8);
+    if (sink.err) return 0.0;
+    int blocks = 256 * 8, iters = 4000;
+    hipLaunchKernelGGL(k_fpmul_bench, dim3(blocks), dim3(256), 0, 0,
+                       sink.as<uint64_t>(), 200, variant);
+    (void)hipDeviceSynchronize();
+    hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0); (void)hipEventCreate(&e1);
+    (void)hipEventRecord(e0, 0);
+    hipLaunchKernelGGL(k_fpmul_bench, dim3(blocks), dim3(256), 0, 0,
+                       sink.as<uint64_t>(), iters, variant);
+    (void)hipEventRecord(e1, 0);
+    if (hipEventSynchronize(e1) != hipSuccess) return 0.0;
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    (void)hipEventDestroy(e0); (void)hipEventDestroy(e1);
+    return (double)blocks * 256.0 * iters * 4.0 / (ms * 1e-3);
+}
+
 /* ---- VALU integer-MAD throughput microbenchmark ----
  * The hot path is wide-integer modular arithmetic: the roofline peak is the
  * device's 64x64->128 multiply-accumulate rate, which no datasheet quotes.
