@@ -1171,17 +1171,33 @@ __global__ void k_g1_table_build(const uint8_t *pks48, g1aff_t *table, int32_t *
  * Each thread sums its strided subset with mixed adds; LDS tree of Jacobian
  * adds reduces to one point (group law is associative => bit-exact). */
 #define MASK_BLOCK 256
+/* full_sum: committee-wide sum (computed once at table build), or NULL.
+ * Dense masks (participation > 1/2 — the FBFT norm is ~90%+) are computed
+ * as full_sum - sum(unset keys): identical group element (and therefore
+ * identical serialized bytes), ~9x less work at 0.9 density. */
 __global__ void __launch_bounds__(MASK_BLOCK)
 k_mask_aggregate(const g1aff_t *table, int n, const uint8_t *bitmaps,
-                 int bm_stride, g1_t *out, int batch) {
+                 int bm_stride, const g1_t *full_sum, g1_t *out, int batch) {
     __shared__ g1_t red[MASK_BLOCK];
+    __shared__ int s_cnt;
     int item = blockIdx.x;
     if (item >= batch) return;
     const uint8_t *bm = bitmaps + (size_t)item * bm_stride;
+    bool complement = false;
+    if (full_sum != nullptr) {
+        if (threadIdx.x == 0) {
+            int cnt = 0;
+            for (int i = 0; i < bm_stride; i++) cnt += __popc(bm[i]);
+            s_cnt = cnt;
+        }
+        __syncthreads();
+        complement = s_cnt > n / 2;
+    }
     g1_t acc;
     g1_set_inf(acc);
     for (int i = threadIdx.x; i < n; i += MASK_BLOCK) {
-        if ((bm[i >> 3] >> (i & 7)) & 1)
+        bool bit = (bm[i >> 3] >> (i & 7)) & 1;
+        if (bit != complement)
             g1_madd(acc, acc, table[i]);
     }
     red[threadIdx.x] = acc;
@@ -1194,7 +1210,16 @@ k_mask_aggregate(const g1aff_t *table, int n, const uint8_t *bitmaps,
         }
         __syncthreads();
     }
-    if (threadIdx.x == 0) out[item] = red[0];
+    if (threadIdx.x == 0) {
+        if (complement) {
+            g1_t nsum, res;
+            g1_neg(nsum, red[0]);
+            g1_add(res, *full_sum, nsum);
+            out[item] = res;
+        } else {
+            out[item] = red[0];
+        }
+    }
 }
 
 __global__ void k_hash_to_g2(const uint8_t *msgs, int mlen, g2_t *out,
@@ -1530,6 +1555,7 @@ extern "C" int hbls_sign_hash(const uint8_t sk32[32], const uint8_t *msg, size_t
 /* ---- committee ---- */
 struct hbls_committee {
     g1aff_t *d_table;
+    g1_t *d_full_sum;   /* committee-wide key sum, for the dense-mask path */
     size_t n;
 };
 extern "C" hbls_committee_t *hbls_committee_build(const uint8_t *pks48, size_t n) {
@@ -1550,10 +1576,33 @@ extern "C" hbls_committee_t *hbls_committee_build(const uint8_t *pks48, size_t n
     hbls_committee_t *c = new hbls_committee_t;
     c->d_table = d_table;
     c->n = n;
+    c->d_full_sum = nullptr;
+    /* committee-wide sum for the dense-mask complement path */
+    g1_t *d_fs = nullptr;
+    if (hipMalloc(&d_fs, sizeof(g1_t)) == hipSuccess) {
+        size_t bm = (n + 7) / 8;
+        DevBuf ones(bm);
+        if (!ones.err) {
+            std::vector<uint8_t> host_ones(bm, 0);
+            for (size_t i = 0; i < n; i++) host_ones[i >> 3] |= 1 << (i & 7);
+            if (hipMemcpy(ones.p, host_ones.data(), bm, hipMemcpyHostToDevice) == hipSuccess) {
+                hipLaunchKernelGGL(k_mask_aggregate, dim3(1), dim3(MASK_BLOCK), 0, 0,
+                                   c->d_table, (int)n, ones.as<uint8_t>(), (int)bm,
+                                   (const g1_t *)nullptr, d_fs, 1);
+                if (hipDeviceSynchronize() == hipSuccess)
+                    c->d_full_sum = d_fs;
+            }
+        }
+        if (c->d_full_sum == nullptr) hipFree(d_fs);
+    }
     return c;
 }
 extern "C" void hbls_committee_free(hbls_committee_t *c) {
-    if (c) { hipFree(c->d_table); delete c; }
+    if (c) {
+        hipFree(c->d_table);
+        if (c->d_full_sum) hipFree(c->d_full_sum);
+        delete c;
+    }
 }
 extern "C" size_t hbls_committee_size(const hbls_committee_t *c) { return c->n; }
 
@@ -1568,7 +1617,7 @@ extern "C" int hbls_mask_aggregate_g1(const hbls_committee_t *c, const uint8_t *
     Timer tm;
     hipLaunchKernelGGL(k_mask_aggregate, dim3(1), dim3(MASK_BLOCK), 0, 0,
                        c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
-                       dout.as<g1_t>(), 1);
+                       c->d_full_sum, dout.as<g1_t>(), 1);
     hipLaunchKernelGGL(k_g1_serialize, dim3(1), dim3(1), 0, 0,
                        dout.as<g1_t>(), dser.as<uint8_t>(), 1);
     tm.stop_and_store();
@@ -1597,7 +1646,7 @@ extern "C" int hbls_batch_agg_verify(const hbls_committee_t *c, const uint8_t *b
     (void)hipEventRecord(ev[0], 0);
     hipLaunchKernelGGL(k_mask_aggregate, dim3((int)batch), dim3(MASK_BLOCK), 0, 0,
                        c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
-                       dagg.as<g1_t>(), (int)batch);
+                       c->d_full_sum, dagg.as<g1_t>(), (int)batch);
     (void)hipEventRecord(ev[1], 0);
     hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
                        dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
@@ -1972,7 +2021,7 @@ extern "C" int hbls_mask_partials(const hbls_committee_t *c, const uint8_t *bitm
     int nb = (int)((batch + 63) / 64);
     hipLaunchKernelGGL(k_mask_aggregate, dim3((int)batch), dim3(MASK_BLOCK), 0, 0,
                        c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
-                       dagg.as<g1_t>(), (int)batch);
+                       c->d_full_sum, dagg.as<g1_t>(), (int)batch);
     hipLaunchKernelGGL(k_g1_serialize, dim3(nb), dim3(64), 0, 0,
                        dagg.as<g1_t>(), dser.as<uint8_t>(), (int)batch);
     tm.stop_and_store();
@@ -2008,7 +2057,7 @@ extern "C" int hbls_batch_agg_verify_partials(
     int nb = (int)((batch + 63) / 64);
     hipLaunchKernelGGL(k_mask_aggregate, dim3((int)batch), dim3(MASK_BLOCK), 0, 0,
                        c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
-                       dagg.as<g1_t>(), (int)batch);
+                       c->d_full_sum, dagg.as<g1_t>(), (int)batch);
     if (n_ext)
         hipLaunchKernelGGL(k_add_partials, dim3(nb), dim3(64), 0, 0,
                            dagg.as<g1_t>(), dext.as<uint8_t>(), (int)n_ext,
