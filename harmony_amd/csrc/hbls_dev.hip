@@ -1170,7 +1170,7 @@ __global__ void k_g1_table_build(const uint8_t *pks48, g1aff_t *table, int32_t *
 /* masked aggregate: one block (256 threads) per batch item.
  * Each thread sums its strided subset with mixed adds; LDS tree of Jacobian
  * adds reduces to one point (group law is associative => bit-exact). */
-#define MASK_BLOCK 256
+#define MASK_BLOCK 64
 /* full_sum: committee-wide sum (computed once at table build), or NULL.
  * Dense masks (participation > 1/2 — the FBFT norm is ~90%+) are computed
  * as full_sum - sum(unset keys): identical group element (and therefore
