@@ -57,7 +57,8 @@ def build_inputs(rank, batch):
         for b in range(hi - lo):
             sk_sums.append(sum(int(sums[b, j]) << (16 * j) for j in range(16)) % pr.R)
     bitmaps = b"".join(bitmaps)
-    msgs = [pr.construct_commit_payload(j, pr.keccak256(b"blk" + j.to_bytes(8, "little")),
+    from oracle import capi
+    msgs = [pr.construct_commit_payload(j, capi.keccak256(b"blk" + j.to_bytes(8, "little")),
                                         j + 1) for j in range(batch)]
     return sks, bitmaps, sk_sums, msgs
 
@@ -96,8 +97,9 @@ def run_config4(args, rank, world, dist):
         for b in range(b1 - b0):
             sk_sums.append(sum(int(sums[b, j]) << (16 * j) for j in range(16)) % pr.R)
     slice_bms = b"".join(slice_bms)
+    from oracle import capi
     msgs = b"".join(pr.construct_commit_payload(
-        j, pr.keccak256(b"blk" + j.to_bytes(8, "little")), j + 1) for j in range(batch))
+        j, capi.keccak256(b"blk" + j.to_bytes(8, "little")), j + 1) for j in range(batch))
     sigs = core.batch_sign(b"".join(pr.fr_serialize(s) for s in sk_sums),
                            msgs, MSG_LEN, batch)
 
