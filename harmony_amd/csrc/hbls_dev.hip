@@ -914,66 +914,97 @@ DEVN void fp12_mul_line(fp12_t &f, const fp2_t &c0, const fp2_t &c3, const fp2_t
     fp6_add(f.c0, t0, vt1);
     f.c1 = tt;
 }
-/* f_{|z|,Q}(P); accumulate into f (caller inits f=1 or continues a product) */
+/* Miller-loop step helpers: line coefficients from the Jacobian formulas
+ * (derivation in DESIGN.md §4), point updated in place. */
+DEVN void ml_dbl_step(g2_t &T, const g1aff_t &Pa, fp2_t &c0, fp2_t &c3, fp2_t &c5) {
+    fp2_t A, B, ZZ, t, t2, newz;
+    fp2_sqr(A, T.x);
+    fp2_sqr(B, T.y);
+    fp2_sqr(ZZ, T.z);
+    fp2_mul(t, A, T.x);
+    fp2_dbl(t2, t); fp2_add(t, t, t2);
+    fp2_dbl(t2, B);
+    fp2_sub(c3, t, t2);                  /* 3X^3 - 2Y^2 */
+    fp2_dbl(t, A); fp2_add(t, t, A);
+    fp2_mul(t, t, ZZ);
+    fp2_mul_fp(t, t, Pa.x);
+    fp2_neg(c5, t);                      /* -3X^2 Z^2 xP */
+    fp2_mul(newz, T.y, T.z);
+    fp2_dbl(newz, newz);
+    fp2_mul(t, newz, ZZ);
+    fp2_mul_fp(t, t, Pa.y);
+    fp2_mul_xi(c0, t);                   /* xi yP 2YZ^3 */
+    g2_dbl(T, T);
+}
+DEVN void ml_add_step(g2_t &T, const g2aff_t &Q, const g1aff_t &Pa,
+                      fp2_t &c0, fp2_t &c3, fp2_t &c5) {
+    fp2_t zz, u2, s2, h, rr, zh, hh, hhh, v, newx, t, t2;
+    fp2_sqr(zz, T.z);
+    fp2_mul(u2, Q.x, zz);
+    fp2_mul(s2, Q.y, zz);
+    fp2_mul(s2, s2, T.z);
+    fp2_sub(h, u2, T.x);
+    fp2_sub(rr, s2, T.y);
+    fp2_mul(zh, T.z, h);
+    fp2_mul_fp(t, zh, Pa.y);
+    fp2_mul_xi(c0, t);                   /* xi yP ZH */
+    fp2_mul(t, rr, Q.x);
+    fp2_mul(t2, Q.y, zh);
+    fp2_sub(c3, t, t2);                  /* r x2 - y2 ZH */
+    fp2_mul_fp(t, rr, Pa.x);
+    fp2_neg(c5, t);                      /* -r xP */
+    fp2_sqr(hh, h);
+    fp2_mul(hhh, hh, h);
+    fp2_mul(v, T.x, hh);
+    fp2_sqr(t, rr);
+    fp2_sub(t, t, hhh);
+    fp2_sub(t, t, v);
+    fp2_sub(newx, t, v);
+    fp2_sub(t, v, newx);
+    fp2_mul(t, rr, t);
+    fp2_mul(t2, T.y, hhh);
+    fp2_sub(T.y, t, t2);
+    T.x = newx;
+    fp2_mul(T.z, T.z, h);
+}
+/* f_{|z|,Q}(P); f must start at 1 */
 DEVN void miller_loop_acc(fp12_t &f, const g2aff_t &Q, const g1aff_t &Pa) {
     g2_t T;
     g2_from_affine(T, Q);
+    fp2_t c0, c3, c5;
     for (int bit = 62; bit >= 0; bit--) {
         fp12_sqr(f, f);
-        fp2_t A, B, ZZ, c0, c3, c5, t, t2, newz;
-        fp2_sqr(A, T.x);
-        fp2_sqr(B, T.y);
-        fp2_sqr(ZZ, T.z);
-        fp2_mul(t, A, T.x);
-        fp2_dbl(t2, t); fp2_add(t, t, t2);
-        fp2_dbl(t2, B);
-        fp2_sub(c3, t, t2);
-        fp2_dbl(t, A); fp2_add(t, t, A);
-        fp2_mul(t, t, ZZ);
-        fp2_mul_fp(t, t, Pa.x);
-        fp2_neg(c5, t);
-        fp2_mul(newz, T.y, T.z);
-        fp2_dbl(newz, newz);
-        fp2_mul(t, newz, ZZ);
-        fp2_mul_fp(t, t, Pa.y);
-        fp2_mul_xi(c0, t);
-        g2_dbl(T, T);
+        ml_dbl_step(T, Pa, c0, c3, c5);
         fp12_mul_line(f, c0, c3, c5);
         if ((BLS_U >> bit) & 1) {
-            fp2_t zz, u2, s2, h, rr, zh, hh, hhh, v, newx;
-            fp2_sqr(zz, T.z);
-            fp2_mul(u2, Q.x, zz);
-            fp2_mul(s2, Q.y, zz);
-            fp2_mul(s2, s2, T.z);
-            fp2_sub(h, u2, T.x);
-            fp2_sub(rr, s2, T.y);
-            fp2_mul(zh, T.z, h);
-            fp2_mul_fp(t, zh, Pa.y);
-            fp2_mul_xi(c0, t);
-            fp2_mul(t, rr, Q.x);
-            fp2_mul(t2, Q.y, zh);
-            fp2_sub(c3, t, t2);
-            fp2_mul_fp(t, rr, Pa.x);
-            fp2_neg(c5, t);
-            fp2_sqr(hh, h);
-            fp2_mul(hhh, hh, h);
-            fp2_mul(v, T.x, hh);
-            fp2_sqr(t, rr);
-            fp2_sub(t, t, hhh);
-            fp2_sub(t, t, v);
-            fp2_sub(newx, t, v);
-            fp2_sub(t, v, newx);
-            fp2_mul(t, rr, t);
-            fp2_mul(t2, T.y, hhh);
-            fp2_sub(T.y, t, t2);
-            T.x = newx;
-            fp2_mul(T.z, T.z, h);
+            ml_add_step(T, Q, Pa, c0, c3, c5);
             fp12_mul_line(f, c0, c3, c5);
         }
     }
 }
-/* Granger-Scott cyclotomic squaring (cyclotomic-subgroup elements only;
- * validated against fp12_sqr by the oracle self-test + GPU parity) */
+/* fused two-pairing Miller loop: f = f_{|z|,Q1}(P1) * f_{|z|,Q2}(P2), one
+ * shared squaring chain (the standard multi-pairing product trick) */
+DEVN void miller_loop2(fp12_t &f, const g2aff_t &Q1, const g1aff_t &P1,
+                       const g2aff_t &Q2, const g1aff_t &P2) {
+    g2_t T1, T2;
+    g2_from_affine(T1, Q1);
+    g2_from_affine(T2, Q2);
+    fp2_t c0, c3, c5;
+    fp12_one(f);
+    for (int bit = 62; bit >= 0; bit--) {
+        fp12_sqr(f, f);
+        ml_dbl_step(T1, P1, c0, c3, c5);
+        fp12_mul_line(f, c0, c3, c5);
+        ml_dbl_step(T2, P2, c0, c3, c5);
+        fp12_mul_line(f, c0, c3, c5);
+        if ((BLS_U >> bit) & 1) {
+            ml_add_step(T1, Q1, P1, c0, c3, c5);
+            fp12_mul_line(f, c0, c3, c5);
+            ml_add_step(T2, Q2, P2, c0, c3, c5);
+            fp12_mul_line(f, c0, c3, c5);
+        }
+    }
+}
 DEV void fp4_sqr_gs(fp2_t &c, fp2_t &d, const fp2_t &a, const fp2_t &b) {
     fp2_t a2, b2, t;
     fp2_sqr(a2, a);
@@ -1052,15 +1083,8 @@ DEVN int verify_pairing(const g1_t &pub, const g2_t &hm, const g2aff_t &sig_aff,
       ba.x = nb.x; ba.y = nb.y; }
     g2_t hmc = hm;
     g2_to_affine(ha, hmc);
-    /* the Miller accumulator's squarings require f to start at 1, so the two
-     * loops run separately and multiply (shared-squaring fusion is a later
-     * optimization) */
-    fp12_t f, f2;
-    fp12_one(f);
-    miller_loop_acc(f, ha, pa);
-    fp12_one(f2);
-    miller_loop_acc(f2, sig_aff, ba);
-    fp12_mul(f, f, f2);
+    fp12_t f;
+    miller_loop2(f, ha, pa, sig_aff, ba);
     fp12_conj(f, f);
     final_exp(f, f);
     return fp12_is_one(f) ? 1 : 0;
