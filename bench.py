@@ -180,7 +180,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=int(os.environ.get("HBLS_BENCH_BATCH", "49152")))
+    ap.add_argument("--batch", type=int, default=int(os.environ.get("HBLS_BENCH_BATCH", "131072")))
     ap.add_argument("--mode", choices=["config2", "config4"], default="config2",
                     help="config2: per-rank replica committees (default, weak scaling); "
                          "config4: one 65536-key committee sharded across ranks with "
