@@ -52,6 +52,10 @@ def _load():
                                             ctypes.POINTER(ctypes.c_int32)]
     lib.hbls_construct_commit_payload.argtypes = [ctypes.c_uint64, ctypes.c_char_p,
                                                   ctypes.c_uint64, ctypes.c_int, ctypes.c_char_p]
+    lib.hbls_batch_seal_verify.argtypes = [
+        ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t,
+        ctypes.c_char_p, ctypes.c_size_t, ctypes.c_size_t,
+        ctypes.POINTER(ctypes.c_int32)]
     lib.hbls_mask_partials.argtypes = [ctypes.c_void_p, ctypes.c_char_p,
                                        ctypes.c_size_t, ctypes.c_char_p]
     lib.hbls_batch_agg_verify_partials.argtypes = [
@@ -227,6 +231,14 @@ class Committee:
         _check(_lib.hbls_batch_agg_verify_partials(
             self._h, bitmaps, ext48s, n_ext, sigs, msgs, mlen, batch, res),
             "batch_agg_verify_partials")
+        return list(res)
+
+    def batch_seal_verify(self, sig_bitmaps: bytes, blob_len: int,
+                          msgs: bytes, mlen: int, batch: int):
+        res = (ctypes.c_int32 * batch)()
+        _check(_lib.hbls_batch_seal_verify(self._h, sig_bitmaps, blob_len,
+                                           msgs, mlen, batch, res),
+               "batch_seal_verify")
         return list(res)
 
     def batch_verify_votes(self, key_idx, sigs: bytes, msgs: bytes, mlen: int):

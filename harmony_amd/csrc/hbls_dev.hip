@@ -117,50 +117,57 @@ DEV void fp_dbl(fp_t &r, const fp_t &x) { fp_add(r, x, x); }
 
 /* 12x32-limb CIOS — every product is one v_mad_u64_u32 (32x32+64); measured
  * 1.29x the 6x64/u128 form on gfx950 (hbls_fpmul_bench_ops A/B).  The u64
- * interface/layout is unchanged; split/repack is ~5% of the mads. */
-DEVN void fp_mul(fp_t &r, const fp_t &x, const fp_t &y) {
-    uint32_t a[12], b[12], t[13];
-#pragma unroll
-    for (int i = 0; i < 6; i++) {
-        a[2 * i] = (uint32_t)x.l[i];
-        a[2 * i + 1] = (uint32_t)(x.l[i] >> 32);
-        b[2 * i] = (uint32_t)y.l[i];
-        b[2 * i + 1] = (uint32_t)(y.l[i] >> 32);
-    }
-#pragma unroll
-    for (int i = 0; i < 13; i++) t[i] = 0;
-    uint32_t t13 = 0;
-    const uint32_t pinv32 = (uint32_t)BLS_P_INV;
-#pragma unroll
-    for (int i = 0; i < 12; i++) {
-        uint64_t acc = 0;
-        uint32_t ai = a[i];
-#pragma unroll
-        for (int j = 0; j < 12; j++) {
-            acc = (uint64_t)ai * b[j] + t[j] + (uint32_t)(acc >> 32);
-            t[j] = (uint32_t)acc;
-        }
-        acc = (uint64_t)t[12] + (uint32_t)(acc >> 32);
-        t[12] = (uint32_t)acc;
-        t13 = (uint32_t)(acc >> 32);
-        uint32_t m = t[0] * pinv32;
-        acc = (uint64_t)m * (uint32_t)BLS_P[0] + t[0];
-#pragma unroll
-        for (int j = 1; j < 12; j++) {
-            uint32_t pj = (uint32_t)(BLS_P[j >> 1] >> ((j & 1) * 32));
-            acc = (uint64_t)m * pj + t[j] + (uint32_t)(acc >> 32);
-            t[j - 1] = (uint32_t)acc;
-        }
-        acc = (uint64_t)t[12] + (uint32_t)(acc >> 32);
-        t[11] = (uint32_t)acc;
-        t[12] = t13 + (uint32_t)(acc >> 32);
-    }
-    uint64_t w[6];
-#pragma unroll
-    for (int i = 0; i < 6; i++)
-        w[i] = (uint64_t)t[2 * i] | ((uint64_t)t[2 * i + 1] << 32);
-    fp_cond_sub_p(r, w, t[12]);
-}
+ * interface/layout is unchanged; split/repack is ~5% of the mads.
+ * The body is shared between the default CALL form (noinline: a fully
+ * inlined build thrashes the instruction cache and ran pathologically
+ * slow) and an inline clone used only inside fp2_mul/fp2_sqr. */
+#define FP_MUL_BODY(r, x, y) do { \
+    uint32_t a_[12], b_[12], t_[13]; \
+    _Pragma("unroll") \
+    for (int i_ = 0; i_ < 6; i_++) { \
+        a_[2 * i_] = (uint32_t)(x).l[i_]; \
+        a_[2 * i_ + 1] = (uint32_t)((x).l[i_] >> 32); \
+        b_[2 * i_] = (uint32_t)(y).l[i_]; \
+        b_[2 * i_ + 1] = (uint32_t)((y).l[i_] >> 32); \
+    } \
+    _Pragma("unroll") \
+    for (int i_ = 0; i_ < 13; i_++) t_[i_] = 0; \
+    uint32_t t13_ = 0; \
+    const uint32_t pinv32_ = (uint32_t)BLS_P_INV; \
+    _Pragma("unroll") \
+    for (int i_ = 0; i_ < 12; i_++) { \
+        uint64_t acc_ = 0; \
+        uint32_t ai_ = a_[i_]; \
+        _Pragma("unroll") \
+        for (int j_ = 0; j_ < 12; j_++) { \
+            acc_ = (uint64_t)ai_ * b_[j_] + t_[j_] + (uint32_t)(acc_ >> 32); \
+            t_[j_] = (uint32_t)acc_; \
+        } \
+        acc_ = (uint64_t)t_[12] + (uint32_t)(acc_ >> 32); \
+        t_[12] = (uint32_t)acc_; \
+        t13_ = (uint32_t)(acc_ >> 32); \
+        uint32_t m_ = t_[0] * pinv32_; \
+        acc_ = (uint64_t)m_ * (uint32_t)BLS_P[0] + t_[0]; \
+        _Pragma("unroll") \
+        for (int j_ = 1; j_ < 12; j_++) { \
+            uint32_t pj_ = (uint32_t)(BLS_P[j_ >> 1] >> ((j_ & 1) * 32)); \
+            acc_ = (uint64_t)m_ * pj_ + t_[j_] + (uint32_t)(acc_ >> 32); \
+            t_[j_ - 1] = (uint32_t)acc_; \
+        } \
+        acc_ = (uint64_t)t_[12] + (uint32_t)(acc_ >> 32); \
+        t_[11] = (uint32_t)acc_; \
+        t_[12] = t13_ + (uint32_t)(acc_ >> 32); \
+    } \
+    uint64_t w_[6]; \
+    _Pragma("unroll") \
+    for (int i_ = 0; i_ < 6; i_++) \
+        w_[i_] = (uint64_t)t_[2 * i_] | ((uint64_t)t_[2 * i_ + 1] << 32); \
+    fp_cond_sub_p(r, w_, t_[12]); \
+} while (0)
+
+DEV void fp_mul_inl(fp_t &r, const fp_t &x, const fp_t &y) { FP_MUL_BODY(r, x, y); }
+DEV void fp_sqr_inl(fp_t &r, const fp_t &x) { FP_MUL_BODY(r, x, x); }
+DEVN void fp_mul(fp_t &r, const fp_t &x, const fp_t &y) { FP_MUL_BODY(r, x, y); }
 DEV void fp_sqr(fp_t &r, const fp_t &x) { fp_mul(r, x, x); }
 
 DEV void fp_one(fp_t &r) {
@@ -196,9 +203,9 @@ DEVN void fp_pow(fp_t &r, const fp_t &a, const uint64_t *e, int n) {
     bool started = false;
     for (int i = n - 1; i >= 0; i--)
         for (int b = 63; b >= 0; b--) {
-            if (started) fp_sqr(acc, acc);
+            if (started) fp_sqr_inl(acc, acc);
             if ((e[i] >> b) & 1) {
-                if (started) fp_mul(acc, acc, a);
+                if (started) fp_mul_inl(acc, acc, a);
                 else { acc = a; started = true; }
             }
         }
@@ -236,11 +243,11 @@ DEV void fp2_conj(fp2_t &r, const fp2_t &x) { r.a = x.a; fp_neg(r.b, x.b); }
 DEV void fp2_dbl(fp2_t &r, const fp2_t &x) { fp2_add(r, x, x); }
 DEVN void fp2_mul(fp2_t &r, const fp2_t &x, const fp2_t &y) {
     fp_t ac, bd, ab, cd, t;
-    fp_mul(ac, x.a, y.a);
-    fp_mul(bd, x.b, y.b);
+    fp_mul_inl(ac, x.a, y.a);
+    fp_mul_inl(bd, x.b, y.b);
     fp_add(ab, x.a, x.b);
     fp_add(cd, y.a, y.b);
-    fp_mul(t, ab, cd);
+    fp_mul_inl(t, ab, cd);
     fp_sub(t, t, ac);
     fp_sub(t, t, bd);
     fp_sub(r.a, ac, bd);
@@ -250,8 +257,8 @@ DEVN void fp2_sqr(fp2_t &r, const fp2_t &x) {
     fp_t s, d, m;
     fp_add(s, x.a, x.b);
     fp_sub(d, x.a, x.b);
-    fp_mul(m, x.a, x.b);
-    fp_mul(r.a, s, d);
+    fp_mul_inl(m, x.a, x.b);
+    fp_mul_inl(r.a, s, d);
     fp_dbl(r.b, m);
 }
 DEV void fp2_mul_fp(fp2_t &r, const fp2_t &x, const fp_t &s) {
@@ -320,24 +327,24 @@ DEV bool g1_is_inf(const g1_t &p) { return fp_is_zero(p.z); }
 DEVN void g1_dbl(g1_t &r, const g1_t &p) {
     if (g1_is_inf(p)) { r = p; return; }
     fp_t A, B, C, D, E, F, t;
-    fp_sqr(A, p.x);
-    fp_sqr(B, p.y);
-    fp_sqr(C, B);
+    fp_sqr_inl(A, p.x);
+    fp_sqr_inl(B, p.y);
+    fp_sqr_inl(C, B);
     fp_add(t, p.x, B);
-    fp_sqr(t, t);
+    fp_sqr_inl(t, t);
     fp_sub(t, t, A);
     fp_sub(t, t, C);
     fp_dbl(D, t);
     fp_dbl(E, A);
     fp_add(E, E, A);
-    fp_sqr(F, E);
+    fp_sqr_inl(F, E);
     fp_t nx, nz;
     fp_sub(nx, F, D);
     fp_sub(nx, nx, D);
-    fp_mul(t, p.y, p.z);
+    fp_mul_inl(t, p.y, p.z);
     fp_dbl(nz, t);
     fp_sub(t, D, nx);
-    fp_mul(t, E, t);
+    fp_mul_inl(t, E, t);
     fp_dbl(C, C); fp_dbl(C, C); fp_dbl(C, C);
     fp_sub(r.y, t, C);
     r.x = nx;
@@ -347,58 +354,58 @@ DEVN void g1_add(g1_t &r, const g1_t &p, const g1_t &q) {
     if (g1_is_inf(p)) { r = q; return; }
     if (g1_is_inf(q)) { r = p; return; }
     fp_t z1z1, z2z2, u1, u2, s1, s2, h, rr, hh, hhh, v, t;
-    fp_sqr(z1z1, p.z);
-    fp_sqr(z2z2, q.z);
-    fp_mul(u1, p.x, z2z2);
-    fp_mul(u2, q.x, z1z1);
-    fp_mul(s1, p.y, q.z); fp_mul(s1, s1, z2z2);
-    fp_mul(s2, q.y, p.z); fp_mul(s2, s2, z1z1);
+    fp_sqr_inl(z1z1, p.z);
+    fp_sqr_inl(z2z2, q.z);
+    fp_mul_inl(u1, p.x, z2z2);
+    fp_mul_inl(u2, q.x, z1z1);
+    fp_mul_inl(s1, p.y, q.z); fp_mul_inl(s1, s1, z2z2);
+    fp_mul_inl(s2, q.y, p.z); fp_mul_inl(s2, s2, z1z1);
     fp_sub(h, u2, u1);
     fp_sub(rr, s2, s1);
     if (fp_is_zero(h)) {
         if (fp_is_zero(rr)) { g1_dbl(r, p); return; }
         g1_set_inf(r); return;
     }
-    fp_sqr(hh, h);
-    fp_mul(hhh, hh, h);
-    fp_mul(v, u1, hh);
-    fp_sqr(t, rr);
+    fp_sqr_inl(hh, h);
+    fp_mul_inl(hhh, hh, h);
+    fp_mul_inl(v, u1, hh);
+    fp_sqr_inl(t, rr);
     fp_sub(t, t, hhh);
     fp_sub(t, t, v);
     fp_sub(r.x, t, v);
     fp_sub(t, v, r.x);
-    fp_mul(t, rr, t);
-    fp_mul(v, s1, hhh);
+    fp_mul_inl(t, rr, t);
+    fp_mul_inl(v, s1, hhh);
     fp_sub(r.y, t, v);
-    fp_mul(t, p.z, q.z);
-    fp_mul(r.z, t, h);
+    fp_mul_inl(t, p.z, q.z);
+    fp_mul_inl(r.z, t, h);
 }
 /* mixed add: r = p + affine q (q never infinity — table keys) */
 DEVN void g1_madd(g1_t &r, const g1_t &p, const g1aff_t &q) {
     if (g1_is_inf(p)) { r.x = q.x; r.y = q.y; fp_one(r.z); return; }
     fp_t z1z1, u2, s2, h, rr, hh, hhh, v, t;
-    fp_sqr(z1z1, p.z);
-    fp_mul(u2, q.x, z1z1);
-    fp_mul(s2, q.y, p.z);
-    fp_mul(s2, s2, z1z1);
+    fp_sqr_inl(z1z1, p.z);
+    fp_mul_inl(u2, q.x, z1z1);
+    fp_mul_inl(s2, q.y, p.z);
+    fp_mul_inl(s2, s2, z1z1);
     fp_sub(h, u2, p.x);
     fp_sub(rr, s2, p.y);
     if (fp_is_zero(h)) {
         if (fp_is_zero(rr)) { g1_dbl(r, p); return; }
         g1_set_inf(r); return;
     }
-    fp_sqr(hh, h);
-    fp_mul(hhh, hh, h);
-    fp_mul(v, p.x, hh);
-    fp_sqr(t, rr);
+    fp_sqr_inl(hh, h);
+    fp_mul_inl(hhh, hh, h);
+    fp_mul_inl(v, p.x, hh);
+    fp_sqr_inl(t, rr);
     fp_sub(t, t, hhh);
     fp_sub(t, t, v);
     fp_sub(r.x, t, v);
     fp_sub(t, v, r.x);
-    fp_mul(t, rr, t);
-    fp_mul(v, p.y, hhh);
+    fp_mul_inl(t, rr, t);
+    fp_mul_inl(v, p.y, hhh);
     fp_sub(r.y, t, v);
-    fp_mul(r.z, p.z, h);
+    fp_mul_inl(r.z, p.z, h);
 }
 DEV void g1_neg(g1_t &r, const g1_t &p) { r.x = p.x; fp_neg(r.y, p.y); r.z = p.z; }
 DEVN void g1_to_affine(g1aff_t &r, const g1_t &p) {
@@ -2194,6 +2201,26 @@ extern "C" int hbls_batch_agg_verify_partials(
     HIP_OK(hipGetLastError());
     HIP_OK(hipMemcpy(results, dres.p, batch * 4, hipMemcpyDeviceToHost));
     return HBLS_OK;
+}
+
+/* sync-path batch seal verification (stagedstreamsync/sig_verify.go:23-59,
+ * legacysync/syncing.go:857): each item is a raw commitSigAndBitmap blob
+ * (96B sig || ceil(n/8)B bitmap, internal/chain/sig.go:22-35) plus its
+ * commit payload; parses on the host (byte split) and runs the batched
+ * aggregate-verify pipeline. */
+extern "C" int hbls_batch_seal_verify(const hbls_committee_t *c,
+                                      const uint8_t *sig_bitmaps, size_t blob_len,
+                                      const uint8_t *msgs, size_t msg_len,
+                                      size_t batch, int32_t *results) {
+    size_t bm = (c->n + 7) / 8;
+    if (blob_len != 96 + bm) return HBLS_ERR_BADINPUT;
+    std::vector<uint8_t> sigs(batch * 96), bms(batch * bm);
+    for (size_t j = 0; j < batch; j++) {
+        memcpy(&sigs[j * 96], sig_bitmaps + j * blob_len, 96);
+        memcpy(&bms[j * bm], sig_bitmaps + j * blob_len + 96, bm);
+    }
+    return hbls_batch_agg_verify(c, bms.data(), sigs.data(), msgs, msg_len,
+                                 batch, results);
 }
 
 /* ConstructCommitPayload (consensus/signature/signature.go:12-24):

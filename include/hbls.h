@@ -155,6 +155,13 @@ int hbls_construct_commit_payload(uint64_t block_num, const uint8_t hash32[32],
 int hbls_parse_commit_sig_bitmap(const uint8_t *payload, size_t len,
                                  uint8_t sig96[96], uint8_t *bitmap, size_t bitmap_cap);
 
+/* sync-path batch seal verification (stagedstreamsync/sig_verify.go:23-59):
+ * items are raw commitSigAndBitmap blobs (96B sig || bitmap) + payloads */
+int hbls_batch_seal_verify(const hbls_committee_t *c,
+                           const uint8_t *sig_bitmaps, size_t blob_len,
+                           const uint8_t *msgs, size_t msg_len,
+                           size_t batch, int32_t *results);
+
 /* batch Keccak-256 (consensus message digests, crypto/hash/hash.go:9-15) */
 int hbls_batch_keccak256(const uint8_t *msgs, size_t msg_len, size_t batch,
                          uint8_t *out32s);
