@@ -171,6 +171,12 @@ def msm_g1(points: bytes, scalars: bytes, n: int) -> bytes:
     return out.raw
 
 
+def g2_aggregate(sigs_cat: bytes, n: int) -> bytes:
+    out = ctypes.create_string_buffer(96)
+    _check(_lib.hbls_g2_aggregate(sigs_cat, n, out), "g2_aggregate")
+    return out.raw
+
+
 def batch_keccak256(msgs: bytes, mlen: int, batch: int) -> bytes:
     out = ctypes.create_string_buffer(32 * batch)
     _check(_lib.hbls_batch_keccak256(msgs, mlen, batch, out), "batch_keccak")

@@ -1460,6 +1460,34 @@ __global__ void k_add_partials(g1_t *aggs, const uint8_t *ext48s, int n_ext,
     ok[i] = 1;
 }
 
+/* sum of n serialized G2 signatures (AggregateSig batch form,
+ * crypto/bls/mask.go:57-64): one 64-thread block, strided deserialize+add,
+ * LDS tree.  flags: 1 ok, 0 bad input. */
+__global__ void __launch_bounds__(64) k_g2_sum(const uint8_t *sigs96, int n,
+                                               uint8_t *out96, int32_t *ok) {
+    __shared__ g2_t red[64];
+    g2_t acc;
+    g2_set_inf(acc);
+    bool good = true;
+    for (int i = threadIdx.x; i < n; i += 64) {
+        g2_t q;
+        if (!g2_deserialize(q, sigs96 + (size_t)i * 96, true)) { good = false; break; }
+        g2_add(acc, acc, q);
+    }
+    if (!good) atomicExch(ok, 0);
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int st = 32; st > 0; st >>= 1) {
+        if (threadIdx.x < st) {
+            g2_t t;
+            g2_add(t, red[threadIdx.x], red[threadIdx.x + st]);
+            red[threadIdx.x] = t;
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0 && *ok) g2_serialize(out96, red[0]);
+}
+
 __global__ void k_keccak(const uint8_t *msgs, int mlen, uint8_t *outs, int batch) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= batch) return;
@@ -2245,6 +2273,27 @@ extern "C" int hbls_parse_commit_sig_bitmap(const uint8_t *payload, size_t len,
     memcpy(sig96, payload, 96);
     memcpy(bitmap, payload + 96, bl);
     return (int)bl;
+}
+
+/* AggregateSig batch form: out = sum of n serialized signatures */
+extern "C" int hbls_g2_aggregate(const uint8_t *sigs96, size_t n, uint8_t out96[96]) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    DevBuf ds(n * 96), dout(96), dok(4);
+    if (ds.err || dout.err || dok.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(ds.p, sigs96, n * 96, hipMemcpyHostToDevice));
+    int32_t one = 1;
+    HIP_OK(hipMemcpy(dok.p, &one, 4, hipMemcpyHostToDevice));
+    Timer tm;
+    hipLaunchKernelGGL(k_g2_sum, dim3(1), dim3(64), 0, 0,
+                       ds.as<uint8_t>(), (int)n, dout.as<uint8_t>(), dok.as<int32_t>());
+    tm.stop_and_store();
+    HIP_OK(hipGetLastError());
+    int32_t ok;
+    HIP_OK(hipMemcpy(&ok, dok.p, 4, hipMemcpyDeviceToHost));
+    if (!ok) return HBLS_ERR_BADINPUT;
+    HIP_OK(hipMemcpy(out96, dout.p, 96, hipMemcpyDeviceToHost));
+    return HBLS_OK;
 }
 
 extern "C" int hbls_batch_keccak256(const uint8_t *msgs, size_t msg_len, size_t batch,

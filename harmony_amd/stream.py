@@ -45,14 +45,19 @@ class StreamVerifier:
         # per-vote verify of the commit payload signature
         msgs = self.payload * batch
         res = self.committee.batch_verify_votes(key_idx, sigs_cat, msgs, len(self.payload))
+        fresh = []
         for j, ok in enumerate(res):
             i = key_idx[j]
             if ok == 1 and not (self.bitmap[i >> 3] >> (i & 7)) & 1:
                 self.bitmap[i >> 3] |= 1 << (i & 7)
-                self.agg_sig = core.g2_add(self.agg_sig, sigs_cat[96 * j:96 * (j + 1)])
+                fresh.append(sigs_cat[96 * j:96 * (j + 1)])
                 self.accepted += 1
             else:
                 self.rejected += 1
+        if fresh:
+            # one batched sum + one add (same group result as per-vote adds)
+            batch_sum = core.g2_aggregate(b"".join(fresh), len(fresh))
+            self.agg_sig = core.g2_add(self.agg_sig, batch_sum)
         self._since_check += batch
         if self._since_check >= self.window:
             self._since_check = 0

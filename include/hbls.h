@@ -162,6 +162,9 @@ int hbls_batch_seal_verify(const hbls_committee_t *c,
                            const uint8_t *msgs, size_t msg_len,
                            size_t batch, int32_t *results);
 
+/* AggregateSig batch form (crypto/bls/mask.go:57-64): out = sum of n sigs */
+int hbls_g2_aggregate(const uint8_t *sigs96, size_t n, uint8_t out96[96]);
+
 /* batch Keccak-256 (consensus message digests, crypto/hash/hash.go:9-15) */
 int hbls_batch_keccak256(const uint8_t *msgs, size_t msg_len, size_t batch,
                          uint8_t *out32s);
