@@ -271,18 +271,6 @@ DEV void fp2_mul_xi(fp2_t &r, const fp2_t &x) {
     fp_add(nb, x.a, x.b);
     r.a = na; r.b = nb;
 }
-DEV void fp2_mul_inl(fp2_t &r, const fp2_t &x, const fp2_t &y) {
-    fp_t ac, bd, ab, cd, t;
-    fp_mul_inl(ac, x.a, y.a);
-    fp_mul_inl(bd, x.b, y.b);
-    fp_add(ab, x.a, x.b);
-    fp_add(cd, y.a, y.b);
-    fp_mul_inl(t, ab, cd);
-    fp_sub(t, t, ac);
-    fp_sub(t, t, bd);
-    fp_sub(r.a, ac, bd);
-    r.b = t;
-}
 DEVN void fp2_inv(fp2_t &r, const fp2_t &x) {
     fp_t n, t, ia, ib;
     fp_sqr(n, x.a);
@@ -756,19 +744,19 @@ DEV void fp6_sub(fp6_t &r, const fp6_t &x, const fp6_t &y) { fp2_sub(r.c0, x.c0,
 DEV void fp6_neg(fp6_t &r, const fp6_t &x) { fp2_neg(r.c0, x.c0); fp2_neg(r.c1, x.c1); fp2_neg(r.c2, x.c2); }
 DEVN void fp6_mul(fp6_t &r, const fp6_t &x, const fp6_t &y) {
     fp2_t t0, t1, t2, s0, s1, tt, r0, r1;
-    fp2_mul_inl(t0, x.c0, y.c0);
-    fp2_mul_inl(t1, x.c1, y.c1);
-    fp2_mul_inl(t2, x.c2, y.c2);
+    fp2_mul(t0, x.c0, y.c0);
+    fp2_mul(t1, x.c1, y.c1);
+    fp2_mul(t2, x.c2, y.c2);
     fp2_add(s0, x.c1, x.c2);
     fp2_add(s1, y.c1, y.c2);
-    fp2_mul_inl(tt, s0, s1);
+    fp2_mul(tt, s0, s1);
     fp2_sub(tt, tt, t1);
     fp2_sub(tt, tt, t2);
     fp2_mul_xi(tt, tt);
     fp2_add(r0, t0, tt);
     fp2_add(s0, x.c0, x.c1);
     fp2_add(s1, y.c0, y.c1);
-    fp2_mul_inl(tt, s0, s1);
+    fp2_mul(tt, s0, s1);
     fp2_sub(tt, tt, t0);
     fp2_sub(tt, tt, t1);
     fp2_t xt2;
@@ -776,7 +764,7 @@ DEVN void fp6_mul(fp6_t &r, const fp6_t &x, const fp6_t &y) {
     fp2_add(r1, tt, xt2);
     fp2_add(s0, x.c0, x.c2);
     fp2_add(s1, y.c0, y.c2);
-    fp2_mul_inl(tt, s0, s1);
+    fp2_mul(tt, s0, s1);
     fp2_sub(tt, tt, t0);
     fp2_sub(tt, tt, t2);
     fp2_add(r.c2, tt, t1);
