@@ -69,3 +69,55 @@ class StreamVerifier:
 
     def final_check(self) -> bool:
         return self.committee.agg_verify(bytes(self.bitmap), self.agg_sig, self.payload)
+
+
+class MultiStreamVerifier:
+    """R concurrent FBFT rounds (consecutive blocks / multiple shards) sharing
+    one committee: pending votes from ALL rounds are verified in ONE
+    batch_verify_votes launch, then folded into each round's aggregate.
+    This is the production shape of the vote pipeline — per-item latency is
+    amortized across rounds in flight, so throughput tracks the batch rate
+    instead of the single-pairing latency."""
+
+    def __init__(self, pks_cat: bytes, n: int, payloads, window: int = 100):
+        self.committee = core.Committee(pks_cat, n)
+        self.n = n
+        self.rounds = [StreamVerifier.__new__(StreamVerifier) for _ in payloads]
+        for sv, payload in zip(self.rounds, payloads):
+            sv.committee = self.committee
+            sv.n = n
+            sv.payload = payload
+            sv.window = window
+            sv.bitmap = bytearray((n + 7) // 8)
+            sv.agg_sig = b"\x00" * 96
+            sv.accepted = sv.rejected = sv.window_checks = 0
+            sv._since_check = 0
+        if len(set(len(p) for p in payloads)) > 1:
+            raise ValueError("payloads must share a length for the batched launch")
+        self.mlen = len(payloads[0])
+
+    def process(self, votes):
+        """votes: list of (round_idx, key_idx, sig96 bytes).  One combined
+        verify launch; bookkeeping per round."""
+        batch = len(votes)
+        key_idx = [v[1] for v in votes]
+        sigs = b"".join(v[2] for v in votes)
+        msgs = b"".join(self.rounds[v[0]].payload for v in votes)
+        res = self.committee.batch_verify_votes(key_idx, sigs, msgs, self.mlen)
+        fresh = {}
+        for (r, i, sig), ok in zip(votes, res):
+            sv = self.rounds[r]
+            if ok == 1 and not (sv.bitmap[i >> 3] >> (i & 7)) & 1:
+                sv.bitmap[i >> 3] |= 1 << (i & 7)
+                fresh.setdefault(r, []).append(sig)
+                sv.accepted += 1
+            else:
+                sv.rejected += 1
+        for r, sigs_r in fresh.items():
+            sv = self.rounds[r]
+            s = core.g2_aggregate(b"".join(sigs_r), len(sigs_r))
+            sv.agg_sig = core.g2_add(sv.agg_sig, s)
+        return res
+
+    def final_check_all(self) -> bool:
+        return all(sv.final_check() for sv in self.rounds)
