@@ -120,4 +120,10 @@ class MultiStreamVerifier:
         return res
 
     def final_check_all(self) -> bool:
-        return all(sv.final_check() for sv in self.rounds)
+        """one batched aggregate-verify across every round in flight"""
+        bitmaps = b"".join(bytes(sv.bitmap) for sv in self.rounds)
+        sigs = b"".join(sv.agg_sig for sv in self.rounds)
+        msgs = b"".join(sv.payload for sv in self.rounds)
+        res = self.committee.batch_agg_verify(bitmaps, sigs, msgs, self.mlen,
+                                              len(self.rounds))
+        return all(r == 1 for r in res)
