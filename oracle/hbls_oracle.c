@@ -359,17 +359,6 @@ static void g1_mul(g1_t *r, const g1_t *p, const uint64_t *k, int n) {
         }
     *r = acc;
 }
-static int g1_eq(const g1_t *p, const g1_t *q) {
-    if (g1_is_inf(p) || g1_is_inf(q)) return g1_is_inf(p) && g1_is_inf(q);
-    /* X1 Z2^2 == X2 Z1^2 && Y1 Z2^3 == Y2 Z1^3 */
-    fp_t z1z1, z2z2, a, b;
-    fp_sqr(&z1z1, &p->z); fp_sqr(&z2z2, &q->z);
-    fp_mul(&a, &p->x, &z2z2); fp_mul(&b, &q->x, &z1z1);
-    if (!fp_eq(&a, &b)) return 0;
-    fp_mul(&a, &p->y, &q->z); fp_mul(&a, &a, &z2z2);
-    fp_mul(&b, &q->y, &p->z); fp_mul(&b, &b, &z1z1);
-    return fp_eq(&a, &b);
-}
 
 /* ================================================================== G2 (Jacobian) */
 typedef struct { fp2_t x, y, z; } g2_t;
@@ -466,13 +455,6 @@ static int g2_eq(const g2_t *p, const g2_t *q) {
     return fp2_eq(&a, &b);
 }
 /* psi endomorphism: (x,y) -> (cx*conj(x), cy*conj(y)) */
-static void g2_psi_aff(g2aff_t *r, const g2aff_t *p) {
-    fp2_t cx, cy, t;
-    memcpy(cx.a.l, BLS_PSI_CX_A, 48); memcpy(cx.b.l, BLS_PSI_CX_B, 48);
-    memcpy(cy.a.l, BLS_PSI_CY_A, 48); memcpy(cy.b.l, BLS_PSI_CY_B, 48);
-    fp2_conj(&t, &p->x); fp2_mul(&r->x, &t, &cx);
-    fp2_conj(&t, &p->y); fp2_mul(&r->y, &t, &cy);
-}
 static void g2_psi_jac(g2_t *r, const g2_t *p) {
     /* x=X/Z^2, y=Y/Z^3 -> (cx*conj(X), cy*conj(Y), conj(Z)): no inversion */
     if (g2_is_inf(p)) { *r = *p; return; }
@@ -549,23 +531,7 @@ static int g2_in_subgroup_fast(const g2_t *p) {
     return g2_eq(&lhs, &rhs);
 }
 
-static int g1_on_curve(const g1aff_t *a) {
-    fp_t l, rr, b1;
-    fp_sqr(&l, &a->y);
-    fp_sqr(&rr, &a->x); fp_mul(&rr, &rr, &a->x);
-    memcpy(b1.l, BLS_B1, 48);
-    fp_add(&rr, &rr, &b1);
-    return fp_eq(&l, &rr);
-}
 static void fp2_b2(fp2_t *b) { memcpy(b->a.l, BLS_B2_A, 48); memcpy(b->b.l, BLS_B2_B, 48); }
-static int g2_on_curve(const g2aff_t *a) {
-    fp2_t l, rr, b2;
-    fp2_sqr(&l, &a->y);
-    fp2_sqr(&rr, &a->x); fp2_mul(&rr, &rr, &a->x);
-    fp2_b2(&b2);
-    fp2_add(&rr, &rr, &b2);
-    return fp2_eq(&l, &rr);
-}
 
 int oracle_g1_deserialize(g1_t *p, const uint8_t in[48], int check_subgroup) {
     if (is_all_zero(in, 48)) { g1_set_inf(p); return 1; }
@@ -755,7 +721,7 @@ static void fp6_mul_v(fp6_t *r, const fp6_t *x) {
     r->c2 = x->c1; r->c1 = x->c0; r->c0 = t;
 }
 static void fp6_inv(fp6_t *r, const fp6_t *x) {
-    fp2_t c0, c1, c2, t, t2;
+    fp2_t c0, c1, c2, t;
     fp2_sqr(&c0, &x->c0);
     fp2_mul(&t, &x->c1, &x->c2);
     fp2_mul_xi(&t, &t);
@@ -999,7 +965,7 @@ static void fp4_sqr(fp2_t *c, fp2_t *d, const fp2_t *a, const fp2_t *b) {
 static void fp12_cyc_sqr(fp12_t *r, const fp12_t *x) {
     const fp2_t *a0 = &x->c0.c0, *a1 = &x->c1.c0, *a2 = &x->c0.c1,
                 *a3 = &x->c1.c1, *a4 = &x->c0.c2, *a5 = &x->c1.c2;
-    fp2_t t00, t03, t01, t04, t02, t05, t;
+    fp2_t t00, t03, t01, t04, t02, t05;
     fp4_sqr(&t00, &t03, a0, a3);   /* (a0 + a3 t) */
     fp4_sqr(&t01, &t04, a1, a4);   /* (a1 + a4 t) */
     fp4_sqr(&t02, &t05, a2, a5);   /* (a2 + a5 t) */
@@ -1007,14 +973,6 @@ static void fp12_cyc_sqr(fp12_t *r, const fp12_t *x) {
        r2 = 3 t01 - 2 a2 ; r5 = 3 t04 + 2 a5   (shifted by the w-multiplication)
        r4 = 3 t02 - 2 a4 ; r1 = 3 xi*t05 + 2 a1 */
     fp12_t out;
-#define GS(dst, tv, av, neg_) do { \
-    fp2_t s; \
-    fp2_sub(&s, (tv), (av)); \
-    if (neg_) fp2_neg(&s, &s); \
-    fp2_dbl(&s, &s); \
-    fp2_add(&s, &s, (tv)); \
-    dst = s; } while (0)
-    /* dst = 2*(t - a) + t  (for minus case)  /  2*(t + a) + t (plus case) */
     fp2_t tmp;
     /* r0 = 2(t00 - a0) + t00 */
     fp2_sub(&tmp, &t00, a0); fp2_dbl(&tmp, &tmp); fp2_add(&out.c0.c0, &tmp, &t00);
@@ -1030,8 +988,6 @@ static void fp12_cyc_sqr(fp12_t *r, const fp12_t *x) {
     fp2_add(&tmp, &t03, a3); fp2_dbl(&tmp, &tmp); fp2_add(&out.c1.c1, &tmp, &t03);
     /* r5 (a5, w^5) = 2(t04 + a5) + t04 */
     fp2_add(&tmp, &t04, a5); fp2_dbl(&tmp, &tmp); fp2_add(&out.c1.c2, &tmp, &t04);
-#undef GS
-    (void)t;
     *r = out;
 }
 /* exp by |z|; cyclotomic squarings (callers apply this only after the easy
@@ -1112,14 +1068,8 @@ int oracle_test_cyc_sqr(void) {
     g2aff_t qa;
     fp_base_point(&base);
     g1_to_affine(&ba, &base);
-    {
-        g2_t q;
-        fp2_t xa, ya;
-        memcpy(xa.a.l, BLS_G2_XA, 48); memcpy(xa.b.l, BLS_G2_XB, 48);
-        memcpy(ya.a.l, BLS_G2_YA, 48); memcpy(ya.b.l, BLS_G2_YB, 48);
-        qa.x = xa; qa.y = ya;
-        (void)q;
-    }
+    memcpy(qa.x.a.l, BLS_G2_XA, 48); memcpy(qa.x.b.l, BLS_G2_XB, 48);
+    memcpy(qa.y.a.l, BLS_G2_YA, 48); memcpy(qa.y.b.l, BLS_G2_YB, 48);
     fp12_t f, easy, t, inv, a, b;
     miller_loop(&f, &qa, &ba);
     fp12_conj(&t, &f);
