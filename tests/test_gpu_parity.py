@@ -288,6 +288,59 @@ def test_config4_sharded_partials(core, capi):
     assert res2[0] == 0 and res2[1:] == [1, 1]
 
 
+def test_deserialize_fuzz_agreement(core, capi):
+    """accept/reject parity on arbitrary byte strings: GPU and oracle must
+    agree exactly on every malformed/mutated candidate (G1 and G2)."""
+    import random
+    rng = random.Random(7)
+    pk = capi.pk_from_sk(sk_bytes(0))
+    sig = capi.sign_hash(sk_bytes(0), pr.synth_msg(0))
+    cands48 = [bytes(rng.randrange(256) for _ in range(48)) for _ in range(24)]
+    cands96 = [bytes(rng.randrange(256) for _ in range(96)) for _ in range(16)]
+    for _ in range(8):
+        b = bytearray(pk)
+        b[rng.randrange(48)] ^= 1 << rng.randrange(8)
+        cands48.append(bytes(b))
+        c = bytearray(sig)
+        c[rng.randrange(96)] ^= 1 << rng.randrange(8)
+        cands96.append(bytes(c))
+    for cand in cands48:
+        assert core.g1_check(cand) == capi.g1_check(cand), cand.hex()
+    for cand in cands96:
+        assert core.g2_check(cand) == capi.g2_check(cand), cand.hex()
+
+
+def test_config4_full_size_65536(core):
+    """config-4 committee size on one GPU: slice-partials additivity and a
+    full aggregate-verify at n=65536 (size-independent identities; the
+    oracle only checks the small-slice path elsewhere)."""
+    n = 65536
+    half = n // 2
+    sks_sum_all = 0
+    sks = []
+    for i in range(n):
+        s = pr.synth_sk(i)
+        sks.append(pr.fr_serialize(s))
+        sks_sum_all += s
+    pks = core.batch_pk_from_sk(b"".join(sks), n)
+    full = core.Committee(pks, n)
+    cA = core.Committee(pks[:48 * half], half)
+    cB = core.Committee(pks[48 * half:], half)
+    # full mask: partial(A) + partial(B) == full-committee aggregate
+    bm_full = bytes([0xFF] * (n // 8))
+    bm_half = bytes([0xFF] * (half // 8))
+    pa = cA.mask_partials(bm_half, 1)
+    pb = cB.mask_partials(bm_half, 1)
+    assert core.g1_add(pa, pb) == full.mask_aggregate(bm_full)
+    # aggregate-verify with every key signing (sig = (sum sk)*H(m))
+    msg = pr.construct_commit_payload(5, pr.synth_msg(5), 6)
+    agg = core.sign_hash(pr.fr_serialize(sks_sum_all % pr.R), msg)
+    assert full.agg_verify(bm_full, agg, msg) is True
+    bad = bytearray(bm_full)
+    bad[100] ^= 0x10
+    assert full.agg_verify(bytes(bad), agg, msg) is False
+
+
 # --------------------------------------------------- config-2 size (4096) identities
 @pytest.fixture(scope="module")
 def committee4096(core):
