@@ -1494,12 +1494,23 @@ __global__ void k_keccak(const uint8_t *msgs, int mlen, uint8_t *outs, int batch
     keccak256_dev(msgs + (size_t)i * mlen, mlen, outs + (size_t)i * 32);
 }
 
+#include "hbls_coop.inc"
+
 /* ================================================================ host layer */
 #include <mutex>
 #include <vector>
 
 static int g_device = -1;
 static int g_fast_cofactor = 1;
+static int g_coop_threshold = -1;   /* -1: read env on first use; 0: disabled */
+static int coop_threshold(void) {
+    if (g_coop_threshold < 0) {
+        const char *e = getenv("HBLS_COOP_THRESHOLD");
+        g_coop_threshold = e ? atoi(e) : 4096;
+    }
+    return g_coop_threshold;
+}
+extern "C" void hbls_set_coop_threshold(int n) { g_coop_threshold = n; }
 static thread_local uint64_t g_last_ns = 0;
 static thread_local uint64_t g_stage_ns[8] = {0};
 static std::once_flag g_init_flag;
@@ -1751,9 +1762,16 @@ extern "C" int hbls_batch_agg_verify(const hbls_committee_t *c, const uint8_t *b
     hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                        dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     (void)hipEventRecord(ev[3], 0);
-    hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0,
-                       dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
-                       dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_verify_coop, dim3(nbc), dim3(64), 0, 0,
+                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+                           dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
+    } else {
+        hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0,
+                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+                           dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
+    }
     (void)hipEventRecord(ev[4], 0);
     HIP_OK(hipEventSynchronize(ev[4]));
     float ms_total = 0, ms = 0;
@@ -2294,6 +2312,20 @@ extern "C" int hbls_g2_aggregate(const uint8_t *sigs96, size_t n, uint8_t out96[
     if (!ok) return HBLS_ERR_BADINPUT;
     HIP_OK(hipMemcpy(out96, dout.p, 96, hipMemcpyDeviceToHost));
     return HBLS_OK;
+}
+
+extern "C" int hbls_coop_selftest(void) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return -100;
+    DevBuf dout(4);
+    if (dout.err) return -101;
+    int32_t zero = 0;
+    HIP_OK(hipMemcpy(dout.p, &zero, 4, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_coop_selftest, dim3(1), dim3(64), 0, 0, dout.as<int32_t>());
+    if (hipDeviceSynchronize() != hipSuccess) return -102;
+    int32_t bad;
+    HIP_OK(hipMemcpy(&bad, dout.p, 4, hipMemcpyDeviceToHost));
+    return bad;
 }
 
 extern "C" int hbls_batch_keccak256(const uint8_t *msgs, size_t msg_len, size_t batch,
