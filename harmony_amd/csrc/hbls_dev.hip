@@ -1815,10 +1815,18 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
                        (int)batch, g_fast_cofactor);
     hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                        dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
-    hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
-                       c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
-                       dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
-                       dres.as<int32_t>(), (int)batch);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_verify_votes_coop, dim3(nbc), dim3(64), 0, 0,
+                           c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
+                           dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
+                           dres.as<int32_t>(), (int)batch);
+    } else {
+        hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
+                           c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
+                           dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
+                           dres.as<int32_t>(), (int)batch);
+    }
     tm.stop_and_store();
     HIP_OK(hipGetLastError());
     HIP_OK(hipMemcpy(results, dres.p, batch * 4, hipMemcpyDeviceToHost));
@@ -2240,9 +2248,16 @@ extern "C" int hbls_batch_agg_verify_partials(
                        (int)batch, g_fast_cofactor);
     hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                        dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
-    hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0,
-                       dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
-                       dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_verify_coop, dim3(nbc), dim3(64), 0, 0,
+                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+                           dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
+    } else {
+        hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0,
+                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+                           dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
+    }
     tm.stop_and_store();
     HIP_OK(hipGetLastError());
     HIP_OK(hipMemcpy(results, dres.p, batch * 4, hipMemcpyDeviceToHost));
