@@ -1755,9 +1755,16 @@ extern "C" int hbls_batch_agg_verify(const hbls_committee_t *c, const uint8_t *b
                        c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
                        c->d_full_sum, dagg.as<g1_t>(), (int)batch);
     (void)hipEventRecord(ev[1], 0);
-    hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
-                       dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
-                       (int)batch, g_fast_cofactor);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_hash_to_g2_coop, dim3(nbc), dim3(64), 0, 0,
+                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+                           (int)batch, g_fast_cofactor);
+    } else {
+        hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
+                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+                           (int)batch, g_fast_cofactor);
+    }
     (void)hipEventRecord(ev[2], 0);
     hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                        dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
@@ -1810,9 +1817,16 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
     HIP_OK(hipMemcpy(dmsg.p, msgs, batch * msg_len, hipMemcpyHostToDevice));
     Timer tm;
     int nb = (int)((batch + 63) / 64);
-    hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
-                       dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
-                       (int)batch, g_fast_cofactor);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_hash_to_g2_coop, dim3(nbc), dim3(64), 0, 0,
+                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+                           (int)batch, g_fast_cofactor);
+    } else {
+        hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
+                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+                           (int)batch, g_fast_cofactor);
+    }
     hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                        dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     if ((int)batch <= coop_threshold()) {
@@ -2243,9 +2257,16 @@ extern "C" int hbls_batch_agg_verify_partials(
         hipLaunchKernelGGL(k_add_partials, dim3(nb), dim3(64), 0, 0,
                            dagg.as<g1_t>(), dext.as<uint8_t>(), (int)n_ext,
                            dpok.as<int32_t>(), (int)batch);
-    hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
-                       dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
-                       (int)batch, g_fast_cofactor);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_hash_to_g2_coop, dim3(nbc), dim3(64), 0, 0,
+                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+                           (int)batch, g_fast_cofactor);
+    } else {
+        hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
+                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+                           (int)batch, g_fast_cofactor);
+    }
     hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                        dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     if ((int)batch <= coop_threshold()) {
