@@ -1766,8 +1766,14 @@ extern "C" int hbls_batch_agg_verify(const hbls_committee_t *c, const uint8_t *b
                            (int)batch, g_fast_cofactor);
     }
     (void)hipEventRecord(ev[2], 0);
-    hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
-                       dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_g2_decompress_coop, dim3(nbc), dim3(64), 0, 0,
+                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    } else {
+        hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
+                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    }
     (void)hipEventRecord(ev[3], 0);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
@@ -1827,8 +1833,14 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
                            dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
                            (int)batch, g_fast_cofactor);
     }
-    hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
-                       dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_g2_decompress_coop, dim3(nbc), dim3(64), 0, 0,
+                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    } else {
+        hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
+                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    }
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
         hipLaunchKernelGGL(k_verify_votes_coop, dim3(nbc), dim3(64), 0, 0,
@@ -2267,8 +2279,14 @@ extern "C" int hbls_batch_agg_verify_partials(
                            dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
                            (int)batch, g_fast_cofactor);
     }
-    hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
-                       dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_g2_decompress_coop, dim3(nbc), dim3(64), 0, 0,
+                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    } else {
+        hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
+                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+    }
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
         hipLaunchKernelGGL(k_verify_coop, dim3(nbc), dim3(64), 0, 0,
