@@ -2368,6 +2368,15 @@ extern "C" int hbls_g2_aggregate(const uint8_t *sigs96, size_t n, uint8_t out96[
     return HBLS_OK;
 }
 
+extern "C" int hbls_hash_edge_count(void) {
+    int32_t v = -1;
+    if (hipMemcpyFromSymbol(&v, HIP_SYMBOL(g_cv_edge_count), 4) != hipSuccess) return -1;
+    return v;
+}
+extern "C" int hbls_hash_edge_reset(void) {
+    int32_t z = 0;
+    return hipMemcpyToSymbol(HIP_SYMBOL(g_cv_edge_count), &z, 4) == hipSuccess ? 1 : 0;
+}
 extern "C" int hbls_coop_selftest(void) {
     int rc = require_gpu();
     if (rc != HBLS_OK) return -100;
