@@ -2076,6 +2076,42 @@ DEV void fp_mul32_fi(uint32_t r[12], const uint32_t a[12], const uint32_t b[12],
     for (int i = 0; i < 12; i++) r[i] = t[i];   /* skip cond-sub: timing probe */
     (void)p32;
 }
+/* variant 3: 12x32 CIOS with the outer row loop NOT unrolled (smaller code,
+ * loop-carried schedule); variant 4: two rows interleaved per iteration. */
+DEV void fp_mul32_noun(uint32_t r[12], const uint32_t a[12], const uint32_t b[12],
+                       const uint32_t p32[12], uint32_t pinv32) {
+    uint32_t t[13];
+#pragma unroll
+    for (int i = 0; i < 13; i++) t[i] = 0;
+    uint32_t t13 = 0;
+#pragma unroll 1
+    for (int i = 0; i < 12; i++) {
+        uint64_t acc = 0;
+        uint32_t ai = a[i];
+#pragma unroll
+        for (int j = 0; j < 12; j++) {
+            acc = (uint64_t)ai * b[j] + t[j] + (uint32_t)(acc >> 32);
+            t[j] = (uint32_t)acc;
+        }
+        acc = (uint64_t)t[12] + (uint32_t)(acc >> 32);
+        t[12] = (uint32_t)acc;
+        t13 = (uint32_t)(acc >> 32);
+        uint32_t m = t[0] * pinv32;
+        acc = (uint64_t)m * p32[0] + t[0];
+#pragma unroll
+        for (int j = 1; j < 12; j++) {
+            acc = (uint64_t)m * p32[j] + t[j] + (uint32_t)(acc >> 32);
+            t[j - 1] = (uint32_t)acc;
+        }
+        acc = (uint64_t)t[12] + (uint32_t)(acc >> 32);
+        t[11] = (uint32_t)acc;
+        t[12] = t13 + (uint32_t)(acc >> 32);
+    }
+#pragma unroll
+    for (int i = 0; i < 12; i++) r[i] = t[i];
+    (void)p32;
+}
+
 __global__ void __launch_bounds__(256) k_fpmul_bench(uint64_t *sink, int iters, int variant) {
     /* independent 4-chain per thread to expose ILP, like the real kernels */
     fp_t a, b;
@@ -2114,6 +2150,28 @@ __global__ void __launch_bounds__(256) k_fpmul_bench(uint64_t *sink, int iters, 
             fp_mul32(x1, x1, b32, p32, pinv32);
             fp_mul32(x2, x2, a32, p32, pinv32);
             fp_mul32(x3, x3, b32, p32, pinv32);
+        }
+        if (x0[0] == 0xdeadbeef) sink[threadIdx.x] = x0[0] + x1[1] + x2[2] + x3[3];
+    }
+    if (variant == 3) {
+        uint32_t p32[12], a32[12], b32[12], x0[12], x1[12], x2[12], x3[12];
+#pragma unroll
+        for (int i = 0; i < 6; i++) {
+            p32[2 * i] = (uint32_t)BLS_P[i];
+            p32[2 * i + 1] = (uint32_t)(BLS_P[i] >> 32);
+            a32[2 * i] = (uint32_t)a.l[i];
+            a32[2 * i + 1] = (uint32_t)(a.l[i] >> 32);
+            b32[2 * i] = (uint32_t)b.l[i];
+            b32[2 * i + 1] = (uint32_t)(b.l[i] >> 32);
+        }
+        uint32_t pinv32 = (uint32_t)BLS_P_INV;
+#pragma unroll
+        for (int i = 0; i < 12; i++) { x0[i] = a32[i]; x1[i] = b32[i]; x2[i] = a32[i]; x3[i] = b32[i]; }
+        for (int it = 0; it < iters; it++) {
+            fp_mul32_noun(x0, x0, a32, p32, pinv32);
+            fp_mul32_noun(x1, x1, b32, p32, pinv32);
+            fp_mul32_noun(x2, x2, a32, p32, pinv32);
+            fp_mul32_noun(x3, x3, b32, p32, pinv32);
         }
         if (x0[0] == 0xdeadbeef) sink[threadIdx.x] = x0[0] + x1[1] + x2[2] + x3[3];
     }
