@@ -98,6 +98,12 @@ def set_g2_cofactor_mode(fast: bool):
     _lib.hbls_set_g2_cofactor_mode(int(fast))
 
 
+def set_coop_threshold(n: int):
+    """batches <= n use the wave-cooperative (latency) kernels; larger use
+    thread-per-item (throughput).  -1 restores the env/default value."""
+    _lib.hbls_set_coop_threshold(n)
+
+
 def pk_from_sk(sk32: bytes) -> bytes:
     out = ctypes.create_string_buffer(48)
     _check(_lib.hbls_pk_from_sk(sk32, out), "pk_from_sk")

@@ -385,3 +385,32 @@ def test_config2_full_size(core, capi, committee4096):
     bad = bytearray(bm)
     bad[5] ^= 1 << 3
     assert gc.agg_verify(bytes(bad), agg, msg) is False
+
+
+def test_coop_scalar_kernel_agreement(core, capi, keys16):
+    """both verify kernels (wave-cooperative and thread-per-item) must return
+    identical verdicts on a mixed accept/reject batch."""
+    sks, pks, n = keys16
+    gc = core.Committee(pks, n)
+    bmlen = (n + 7) // 8
+    batch = 5
+    bitmaps, sigs, msgs = b"", b"", b""
+    for j in range(batch):
+        msg = pr.construct_commit_payload(j, pr.synth_msg(j + 50), j)
+        signers = [i for i in range(n) if (i + j) % 2 == 0]
+        bm = bytearray(bmlen)
+        for i in signers:
+            bm[i >> 3] |= 1 << (i & 7)
+        use = signers if j != 3 else signers[:-1]   # item 3 rejects
+        sk_sum = sum(pr.synth_sk(i) for i in use) % pr.R
+        sigs += capi.sign_hash(pr.fr_serialize(sk_sum), msg)
+        bitmaps += bytes(bm)
+        msgs += msg
+    core.set_coop_threshold(10**9)
+    try:
+        coop = gc.batch_agg_verify(bitmaps, sigs, msgs, 48, batch)
+        core.set_coop_threshold(0)
+        scal = gc.batch_agg_verify(bitmaps, sigs, msgs, 48, batch)
+    finally:
+        core.set_coop_threshold(-1)
+    assert coop == scal == [1, 1, 1, 0, 1]
