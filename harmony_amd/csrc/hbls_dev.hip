@@ -2112,6 +2112,64 @@ DEV void fp_mul32_noun(uint32_t r[12], const uint32_t a[12], const uint32_t b[12
     (void)p32;
 }
 
+/* variant 5: comba (product-scanning) interleaved Montgomery — every product
+ * chains into ONE 64-bit accumulator via mad, with a carry-overflow counter
+ * (96-bit column accumulator); no per-limb hi/lo pair shuffling. */
+DEV void fp_mul32_comba(uint32_t r[13], const uint32_t a[12], const uint32_t b[12],
+                        const uint32_t p32[12], uint32_t pinv32) {
+    uint64_t acc = 0;
+    uint32_t extra = 0;
+    uint32_t m[12];
+#pragma unroll
+    for (int k = 0; k < 12; k++) {
+#pragma unroll
+        for (int i = 0; i < 12; i++) {
+            if (i <= k) {
+                uint64_t nacc = acc + (uint64_t)a[i] * b[k - i];
+                extra += (nacc < acc);
+                acc = nacc;
+            }
+        }
+#pragma unroll
+        for (int i = 0; i < 12; i++) {
+            if (i < k) {
+                uint64_t nacc = acc + (uint64_t)m[i] * p32[k - i];
+                extra += (nacc < acc);
+                acc = nacc;
+            }
+        }
+        uint32_t mk = (uint32_t)acc * pinv32;
+        m[k] = mk;
+        {
+            uint64_t nacc = acc + (uint64_t)mk * p32[0];
+            extra += (nacc < acc);
+            acc = nacc;
+        }
+        acc = (acc >> 32) | ((uint64_t)extra << 32);
+        extra = 0;
+    }
+#pragma unroll
+    for (int k = 12; k < 24; k++) {
+#pragma unroll
+        for (int i = 0; i < 12; i++) {
+            if (i >= k - 11) {
+                if (k - i < 12) {
+                    uint64_t nacc = acc + (uint64_t)a[i] * b[k - i];
+                    extra += (nacc < acc);
+                    acc = nacc;
+                    nacc = acc + (uint64_t)m[i] * p32[k - i];
+                    extra += (nacc < acc);
+                    acc = nacc;
+                }
+            }
+        }
+        r[k - 12] = (uint32_t)acc;
+        acc = (acc >> 32) | ((uint64_t)extra << 32);
+        extra = 0;
+    }
+    r[12] = (uint32_t)acc;
+}
+
 __global__ void __launch_bounds__(256) k_fpmul_bench(uint64_t *sink, int iters, int variant) {
     /* independent 4-chain per thread to expose ILP, like the real kernels */
     fp_t a, b;
@@ -2172,6 +2230,29 @@ __global__ void __launch_bounds__(256) k_fpmul_bench(uint64_t *sink, int iters, 
             fp_mul32_noun(x1, x1, b32, p32, pinv32);
             fp_mul32_noun(x2, x2, a32, p32, pinv32);
             fp_mul32_noun(x3, x3, b32, p32, pinv32);
+        }
+        if (x0[0] == 0xdeadbeef) sink[threadIdx.x] = x0[0] + x1[1] + x2[2] + x3[3];
+    }
+    if (variant == 5) {
+        uint32_t p32[12], a32[12], b32[12], x0[13], x1[13], x2[13], x3[13];
+#pragma unroll
+        for (int i = 0; i < 6; i++) {
+            p32[2 * i] = (uint32_t)BLS_P[i];
+            p32[2 * i + 1] = (uint32_t)(BLS_P[i] >> 32);
+            a32[2 * i] = (uint32_t)a.l[i];
+            a32[2 * i + 1] = (uint32_t)(a.l[i] >> 32);
+            b32[2 * i] = (uint32_t)b.l[i];
+            b32[2 * i + 1] = (uint32_t)(b.l[i] >> 32);
+        }
+        uint32_t pinv32 = (uint32_t)BLS_P_INV;
+#pragma unroll
+        for (int i = 0; i < 12; i++) { x0[i] = a32[i]; x1[i] = b32[i]; x2[i] = a32[i]; x3[i] = b32[i]; }
+        x0[12] = x1[12] = x2[12] = x3[12] = 0;
+        for (int it = 0; it < iters; it++) {
+            fp_mul32_comba(x0, x0, a32, p32, pinv32);
+            fp_mul32_comba(x1, x1, b32, p32, pinv32);
+            fp_mul32_comba(x2, x2, a32, p32, pinv32);
+            fp_mul32_comba(x3, x3, b32, p32, pinv32);
         }
         if (x0[0] == 0xdeadbeef) sink[threadIdx.x] = x0[0] + x1[1] + x2[2] + x3[3];
     }
