@@ -1272,7 +1272,7 @@ int oracle_agg_verify(const uint8_t *pks48cat, const uint8_t *bitmap, size_t n,
  * reference's LRU-cached PublicKeyWrapper.Object, crypto/bls/bls.go:30-33,
  * mask.go:13-15); mask-sum then uses mixed (Jacobian+affine) adds. */
 #include <stdlib.h>
-typedef struct { g1aff_t *pts; size_t n; } committee_t;
+typedef struct { g1aff_t *pts; size_t n; g1_t full_sum; } committee_t;
 
 void *oracle_committee_build(const uint8_t *pks48cat, size_t n) {
     committee_t *c = (committee_t *)malloc(sizeof(committee_t));
@@ -1285,6 +1285,18 @@ void *oracle_committee_build(const uint8_t *pks48cat, size_t n) {
             return NULL;
         }
         g1_to_affine(&c->pts[i], &p);   /* keys are never infinity in practice */
+    }
+    /* committee-wide sum, for the dense-mask complement path (same trick as
+     * the GPU mask kernel: group-exact, so serialized results are identical) */
+    g1_set_inf(&c->full_sum);
+    {
+        g1_t acc; g1_set_inf(&acc);
+        for (size_t i = 0; i < n; i++) {
+            /* forward-declared below */
+            extern void oracle_g1_madd_fwd(g1_t *r, const g1_t *p, const g1aff_t *q);
+            oracle_g1_madd_fwd(&acc, &acc, &c->pts[i]);
+        }
+        c->full_sum = acc;
     }
     return c;
 }
@@ -1320,15 +1332,33 @@ static void g1_madd(g1_t *r, const g1_t *p, const g1aff_t *q) {
     fp_mul(&t, &p->z, &h);
     r->z = t;
 }
+void oracle_g1_madd_fwd(g1_t *r, const g1_t *p, const g1aff_t *q) { g1_madd(r, p, q); }
+
 /* one aggregate-verify against a prebuilt table (the north-star unit) */
+static void masked_sum_tab(const committee_t *c, const uint8_t *bitmap, g1_t *out) {
+    size_t cnt = 0;
+    for (size_t i = 0; i < c->n; i++)
+        cnt += (bitmap[i >> 3] >> (i & 7)) & 1;
+    int complement = cnt > c->n / 2;
+    g1_t acc;
+    g1_set_inf(&acc);
+    for (size_t i = 0; i < c->n; i++) {
+        int bit = (bitmap[i >> 3] >> (i & 7)) & 1;
+        if (bit != complement)
+            g1_madd(&acc, &acc, &c->pts[i]);
+    }
+    if (complement) {
+        g1_neg(&acc, &acc);
+        g1_add(out, &c->full_sum, &acc);
+    } else {
+        *out = acc;
+    }
+}
 int oracle_agg_verify_tab(const void *h, const uint8_t *bitmap,
                           const uint8_t sig96[96], const uint8_t *msg, size_t mlen) {
     const committee_t *c = (const committee_t *)h;
     g1_t acc;
-    g1_set_inf(&acc);
-    for (size_t i = 0; i < c->n; i++)
-        if ((bitmap[i >> 3] >> (i & 7)) & 1)
-            g1_madd(&acc, &acc, &c->pts[i]);
+    masked_sum_tab(c, bitmap, &acc);
     uint8_t agg48[48];
     oracle_g1_serialize(agg48, &acc);
     return oracle_verify_hash(agg48, sig96, msg, mlen);
@@ -1350,10 +1380,7 @@ int oracle_batch_agg_verify_tab(const void *h, const uint8_t *bitmaps,
 int oracle_mask_aggregate_tab(const void *h, const uint8_t *bitmap, uint8_t out48[48]) {
     const committee_t *c = (const committee_t *)h;
     g1_t acc;
-    g1_set_inf(&acc);
-    for (size_t i = 0; i < c->n; i++)
-        if ((bitmap[i >> 3] >> (i & 7)) & 1)
-            g1_madd(&acc, &acc, &c->pts[i]);
+    masked_sum_tab(c, bitmap, &acc);
     oracle_g1_serialize(out48, &acc);
     return 1;
 }
