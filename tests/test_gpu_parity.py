@@ -414,3 +414,31 @@ def test_coop_scalar_kernel_agreement(core, capi, keys16):
     finally:
         core.set_coop_threshold(-1)
     assert coop == scal == [1, 1, 1, 0, 1]
+
+
+def test_deserialize_fuzz_extended(core, capi):
+    """larger accept/reject fuzz sweep (mutations of valid points, random
+    bytes, flag-bit flips) — GPU and oracle must agree on every candidate."""
+    import random
+    rng = random.Random(20260915)
+    pk = capi.pk_from_sk(sk_bytes(3))
+    sig = capi.sign_hash(sk_bytes(3), pr.synth_msg(3))
+    c48, c96 = [], []
+    for _ in range(120):
+        c48.append(bytes(rng.randrange(256) for _ in range(48)))
+        b = bytearray(pk)
+        for _ in range(rng.randrange(1, 3)):
+            b[rng.randrange(48)] ^= 1 << rng.randrange(8)
+        c48.append(bytes(b))
+    for _ in range(60):
+        c96.append(bytes(rng.randrange(256) for _ in range(96)))
+        b = bytearray(sig)
+        for _ in range(rng.randrange(1, 3)):
+            b[rng.randrange(96)] ^= 1 << rng.randrange(8)
+        c96.append(bytes(b))
+    # flag-bit-only flips (top bit of last byte): parity selection path
+    b = bytearray(pk); b[47] ^= 0x80; c48.append(bytes(b))
+    b = bytearray(sig); b[95] ^= 0x80; c96.append(bytes(b))
+    mism48 = [c.hex() for c in c48 if core.g1_check(c) != capi.g1_check(c)]
+    mism96 = [c.hex() for c in c96 if core.g2_check(c) != capi.g2_check(c)]
+    assert mism48 == [] and mism96 == []

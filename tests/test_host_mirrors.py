@@ -105,3 +105,37 @@ def test_commit_payload_python_mirror():
     p48 = pr.construct_commit_payload(0x1122334455667788, h, 7, staking=True)
     assert p40 == bytes.fromhex("8877665544332211") + h
     assert p48 == p40 + (7).to_bytes(8, "little")
+
+
+def test_core_codec_helpers():
+    """hbls_construct_commit_payload / hbls_parse_commit_sig_bitmap are host
+    byte code in libhbls.so: loadable and byte-exact without a GPU."""
+    from harmony_amd import core
+    from oracle import pyref as pr
+    h = pr.synth_msg(4)
+    assert core.construct_commit_payload(9, h, 3, staking=True) == \
+        pr.construct_commit_payload(9, h, 3, staking=True)
+    assert core.construct_commit_payload(9, h, 3, staking=False) == \
+        pr.construct_commit_payload(9, h, 3, staking=False)
+    sig = bytes(range(96))
+    bm = b"\xde\xad\xbe"
+    s2, b2 = core.parse_commit_sig_bitmap(sig + bm)
+    assert s2 == sig and b2 == bm
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        core.parse_commit_sig_bitmap(b"\x00" * 95)
+
+
+def test_abi_symbols_exported():
+    """every entry point declared in include/hbls.h resolves in libhbls.so
+    (no compute — loadability check, runs without a GPU)."""
+    import ctypes
+    import re
+    from harmony_amd import core
+    hdr = open(os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "include", "hbls.h")).read()
+    names = re.findall(r"^\s*(?:int|void|double|size_t|uint64_t|const char \*|hbls_committee_t \*)\s*(hbls_\w+)\s*\(",
+                       hdr, re.M)
+    assert len(names) >= 25
+    missing = [n for n in set(names) if not hasattr(core._lib, n)]
+    assert missing == []
