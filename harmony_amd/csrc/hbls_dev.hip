@@ -1506,7 +1506,7 @@ static int g_coop_threshold = -1;   /* -1: read env on first use; 0: disabled */
 static int coop_threshold(void) {
     if (g_coop_threshold < 0) {
         const char *e = getenv("HBLS_COOP_THRESHOLD");
-        g_coop_threshold = e ? atoi(e) : 4096;
+        g_coop_threshold = e ? atoi(e) : 8192;
     }
     return g_coop_threshold;
 }
