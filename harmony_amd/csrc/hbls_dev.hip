@@ -165,9 +165,45 @@ DEV void fp_dbl(fp_t &r, const fp_t &x) { fp_add(r, x, x); }
     fp_cond_sub_p(r, w_, t_[12]); \
 } while (0)
 
+/* Hand-allocated asm CIOS core (generated: gen_fp_mul_asm.py) — same
+ * semantics as FP_MUL_BODY, 938 instrs vs the compiler's 1415 (the pinned
+ * t-register pairs let each v_mad_u64_u32 write (limb, carry) directly,
+ * killing the pair-shuffle movs). */
+#include "fp_mul_asm.inc"
+#define FP_MUL_ASM_BODY(r, x, y) do { \
+    uint32_t a_[12], b_[12], t_[13]; \
+    _Pragma("unroll") \
+    for (int i_ = 0; i_ < 6; i_++) { \
+        a_[2 * i_] = (uint32_t)(x).l[i_]; \
+        a_[2 * i_ + 1] = (uint32_t)((x).l[i_] >> 32); \
+        b_[2 * i_] = (uint32_t)(y).l[i_]; \
+        b_[2 * i_ + 1] = (uint32_t)((y).l[i_] >> 32); \
+    } \
+    fp_mul_asm_core(t_, a_, b_); \
+    uint64_t w_[6]; \
+    _Pragma("unroll") \
+    for (int i_ = 0; i_ < 6; i_++) \
+        w_[i_] = (uint64_t)t_[2 * i_] | ((uint64_t)t_[2 * i_ + 1] << 32); \
+    fp_cond_sub_p(r, w_, t_[12]); \
+} while (0)
+
+/* Measured verdict (see ROUND2_NOTES.md): the asm core is bit-exact and
+ * cuts issue count 1415->1158, but fp_mul is BOUND BY v_mad_u64_u32
+ * EXECUTION, not issue — +4.7% on the called-form µbench, +-0% end to end
+ * (inline sites regress: register pins cost shuffle movs).  Default stays
+ * the compiler body; -DHBLS_ASM_FPMUL switches the called form to asm for
+ * future A/Bs. */
+#ifdef HBLS_ASM_FPMUL
+DEV void fp_mul_inl(fp_t &r, const fp_t &x, const fp_t &y) { FP_MUL_BODY(r, x, y); }
+DEV void fp_sqr_inl(fp_t &r, const fp_t &x) { FP_MUL_BODY(r, x, x); }
+DEVN void fp_mul(fp_t &r, const fp_t &x, const fp_t &y) { FP_MUL_ASM_BODY(r, x, y); }
+#else
 DEV void fp_mul_inl(fp_t &r, const fp_t &x, const fp_t &y) { FP_MUL_BODY(r, x, y); }
 DEV void fp_sqr_inl(fp_t &r, const fp_t &x) { FP_MUL_BODY(r, x, x); }
 DEVN void fp_mul(fp_t &r, const fp_t &x, const fp_t &y) { FP_MUL_BODY(r, x, y); }
+#endif
+DEVN void fp_mul_asm(fp_t &r, const fp_t &x, const fp_t &y) { FP_MUL_ASM_BODY(r, x, y); }
+DEVN void fp_mul_c(fp_t &r, const fp_t &x, const fp_t &y) { FP_MUL_BODY(r, x, y); }
 DEV void fp_sqr(fp_t &r, const fp_t &x) { fp_mul(r, x, x); }
 
 DEV void fp_one(fp_t &r) {
@@ -2183,13 +2219,13 @@ __global__ void __launch_bounds__(256) k_fpmul_bench(uint64_t *sink, int iters, 
     if (variant == 0) {
         fp_t x0 = a, x1 = b, x2 = a, x3 = b;
         for (int it = 0; it < iters; it++) {
-            fp_mul(x0, x0, a);
-            fp_mul(x1, x1, b);
-            fp_mul(x2, x2, a);
-            fp_mul(x3, x3, b);
+            fp_mul_c(x0, x0, a);
+            fp_mul_c(x1, x1, b);
+            fp_mul_c(x2, x2, a);
+            fp_mul_c(x3, x3, b);
         }
         if (x0.l[0] == 0xdeadbeef) sink[threadIdx.x] = x0.l[0] + x1.l[1] + x2.l[2] + x3.l[3];
-    } else {
+    } else if (variant == 1) {
         uint32_t p32[12], a32[12], b32[12], x0[12], x1[12], x2[12], x3[12];
 #pragma unroll
         for (int i = 0; i < 6; i++) {
@@ -2210,6 +2246,16 @@ __global__ void __launch_bounds__(256) k_fpmul_bench(uint64_t *sink, int iters, 
             fp_mul32(x3, x3, b32, p32, pinv32);
         }
         if (x0[0] == 0xdeadbeef) sink[threadIdx.x] = x0[0] + x1[1] + x2[2] + x3[3];
+    }
+    if (variant == 6) {
+        fp_t x0 = a, x1 = b, x2 = a, x3 = b;
+        for (int it = 0; it < iters; it++) {
+            fp_mul_asm(x0, x0, a);
+            fp_mul_asm(x1, x1, b);
+            fp_mul_asm(x2, x2, a);
+            fp_mul_asm(x3, x3, b);
+        }
+        if (x0.l[0] == 0xdeadbeef) sink[threadIdx.x] = x0.l[0] + x1.l[1] + x2.l[2] + x3.l[3];
     }
     if (variant == 3) {
         uint32_t p32[12], a32[12], b32[12], x0[12], x1[12], x2[12], x3[12];
@@ -2279,6 +2325,47 @@ __global__ void __launch_bounds__(256) k_fpmul_bench(uint64_t *sink, int iters, 
         if (x0[0] == 0xdeadbeef) sink[threadIdx.x] = x0[0] + x1[1] + x2[2] + x3[3];
     }
 }
+/* asm-vs-C fp_mul equivalence: each thread runs a dependent chain through
+ * both bodies on xorshift-derived inputs (< p via top-limb mask) and counts
+ * bitwise mismatches at every step. */
+__global__ void k_fpmul_asm_check(uint32_t *mism, int iters) {
+    uint64_t s = 0x9e3779b97f4a7c15ULL * (blockIdx.x * 256 + threadIdx.x + 1);
+    fp_t a, b;
+#pragma unroll
+    for (int i = 0; i < 6; i++) {
+        s ^= s << 13; s ^= s >> 7; s ^= s << 17; a.l[i] = s;
+        s ^= s << 13; s ^= s >> 7; s ^= s << 17; b.l[i] = s;
+    }
+    a.l[5] &= 0x19999999ffffffffULL;   /* < p: top 32-bit limb < 0x1a0111ea */
+    b.l[5] &= 0x19999999ffffffffULL;
+    fp_t xc = a, xa = a;
+    uint32_t bad = 0;
+    for (int it = 0; it < iters; it++) {
+        fp_mul_c(xc, xc, b);
+        fp_mul_asm(xa, xa, b);
+#pragma unroll
+        for (int i = 0; i < 6; i++) bad += (xc.l[i] != xa.l[i]);
+        fp_mul_c(xc, xc, xc);
+        fp_mul_asm(xa, xa, xa);
+#pragma unroll
+        for (int i = 0; i < 6; i++) bad += (xc.l[i] != xa.l[i]);
+    }
+    if (bad) atomicAdd(mism, bad);
+}
+/* returns mismatch count (-1 on launch failure); 0 = asm core bit-exact */
+extern "C" long long hbls_fpmul_asm_check(void) {
+    if (require_gpu() != HBLS_OK) return -1;
+    DevBuf mism(4);
+    if (mism.err) return -1;
+    (void)hipMemset(mism.p, 0, 4);
+    hipLaunchKernelGGL(k_fpmul_asm_check, dim3(512), dim3(256), 0, 0,
+                       mism.as<uint32_t>(), 64);
+    if (hipDeviceSynchronize() != hipSuccess) return -1;
+    uint32_t h = 0;
+    (void)hipMemcpy(&h, mism.p, 4, hipMemcpyDeviceToHost);
+    return (long long)h;
+}
+
 /* measured fp_mul throughput (muls/s) for the given variant */
 extern "C" double hbls_fpmul_bench_ops(int variant) {
     if (require_gpu() != HBLS_OK) return 0.0;
