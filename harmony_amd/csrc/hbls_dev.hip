@@ -1238,27 +1238,31 @@ __global__ void k_g1_table_build(const uint8_t *pks48, g1aff_t *table, int32_t *
  * Each thread sums its strided subset with mixed adds; LDS tree of Jacobian
  * adds reduces to one point (group law is associative => bit-exact). */
 #define MASK_BLOCK 64
-#define MASK_SUBS 4            /* items per 64-thread block */
-#define MASK_LANES 16          /* lanes per item */
-#define MASK_IDX_CAP 1024      /* compacted minority-side capacity per item */
-/* Masked committee sum, one wave per 4 items (16 lanes each).
+#define MASK_SUBS 4            /* items per block */
+#define MASK_LANES 16          /* lanes per item (small-committee instance) */
+#define MASK_IDX_CAP 1024      /* segment length = compacted capacity, bits */
+/* Masked committee sum, MASK_SUBS items per block, LANES lanes per item
+ * (16 for small committees; 64 — a full wave per item — when n >= 16384,
+ * where 16 lanes x batch leaves most of the chip idle).
  * 1) popcount the bitmap; if participation > 1/2 and the committee full-sum
  *    is available, work on the COMPLEMENT side (full - sum(unset)): the
  *    FBFT norm is ~90% participation, so the minority side is ~10% of keys.
- * 2) compact the minority-side indices into LDS, so the point additions run
- *    with every lane active (a strided bit-test loop at 10% density leaves
- *    ~90% of each wave idle — measured 45x off the chip's fp_mul rate).
- * 3) strided fallback when the minority side exceeds the LDS capacity.
+ * 2) walk the bitmap in 1024-bit segments, compacting each segment's
+ *    minority-side indices into LDS word-by-word (one atomic per nonzero
+ *    word), so the point additions run with every lane active — the
+ *    strided bit-test form at 10-33% minority density leaves most of each
+ *    wave idle (measured 45x off the chip's fp_mul rate at 10%).
  * EC addition is associative: any order/tree is bit-exact after
  * normalization (parity tests vs the oracle's sequential loop). */
-__global__ void __launch_bounds__(MASK_BLOCK)
+template <int LANES>
+__global__ void __launch_bounds__(LANES * MASK_SUBS)
 k_mask_aggregate(const g1aff_t *table, int n, const uint8_t *bitmaps,
                  int bm_stride, const g1_t *full_sum, g1_t *out, int batch) {
-    __shared__ g1_t red[MASK_SUBS][MASK_LANES];
-    __shared__ uint16_t idx[MASK_SUBS][MASK_IDX_CAP];
+    __shared__ g1_t red[MASK_SUBS][LANES];
+    __shared__ uint32_t idx[MASK_SUBS][MASK_IDX_CAP];
     __shared__ int cnt[MASK_SUBS];
-    const int sub = threadIdx.x / MASK_LANES;
-    const int lane = threadIdx.x % MASK_LANES;
+    const int sub = threadIdx.x / LANES;
+    const int lane = threadIdx.x % LANES;
     const int item = blockIdx.x * MASK_SUBS + sub;
     const bool active = item < batch;
     const uint8_t *bm = active ? bitmaps + (size_t)item * bm_stride : nullptr;
@@ -1267,45 +1271,53 @@ k_mask_aggregate(const g1aff_t *table, int n, const uint8_t *bitmaps,
     __syncthreads();
     if (active) {
         int c = 0;
-        for (int i = lane; i < bm_stride; i += MASK_LANES) c += __popc(bm[i]);
+        for (int i = lane; i < bm_stride; i += LANES) c += __popc(bm[i]);
         if (c) atomicAdd(&cnt[sub], c);
     }
     __syncthreads();
     const int set_count = cnt[sub];
     const bool complement = active && full_sum != nullptr && set_count > n / 2;
-    const int minority = complement ? n - set_count : set_count;
-    const bool compacted = active && minority <= MASK_IDX_CAP && n < 65536;
-    __syncthreads();
-    if (lane == 0) cnt[sub] = 0;
-    __syncthreads();
-    if (compacted) {
-        for (int i = lane; i < n; i += MASK_LANES) {
-            bool bit = (bm[i >> 3] >> (i & 7)) & 1;
-            if (bit != complement) {
-                int pos = atomicAdd(&cnt[sub], 1);
-                idx[sub][pos] = (uint16_t)i;
-            }
-        }
-    }
-    __syncthreads();
     g1_t acc;
     g1_set_inf(acc);
-    if (active) {
-        if (compacted) {
-            const int m = cnt[sub];
-            for (int k = lane; k < m; k += MASK_LANES)
-                g1_madd(acc, acc, table[idx[sub][k]]);
-        } else {
-            for (int i = lane; i < n; i += MASK_LANES) {
-                bool bit = (bm[i >> 3] >> (i & 7)) & 1;
-                if (bit != complement)
-                    g1_madd(acc, acc, table[i]);
+    for (int base = 0; base < n; base += MASK_IDX_CAP) {
+        __syncthreads();
+        if (lane == 0) cnt[sub] = 0;
+        __syncthreads();
+        if (active) {
+            const int end = base + MASK_IDX_CAP < n ? base + MASK_IDX_CAP : n;
+            for (int w = base / 32 + lane; w * 32 < end; w += LANES) {
+                uint32_t word = 0;
+#pragma unroll
+                for (int k = 0; k < 4; k++) {
+                    int byte = w * 4 + k;
+                    if (byte < bm_stride) word |= (uint32_t)bm[byte] << (8 * k);
+                }
+                if (complement) word = ~word;
+                const int lo = w * 32;
+                /* clip past n: bitmap padding must not enter the complement */
+                if (n - lo < 32) word &= (1u << (n - lo)) - 1;
+                int c = __popc(word);
+                if (c) {
+                    int pos = atomicAdd(&cnt[sub], c);
+                    while (word) {
+                        int b = __ffs(word) - 1;
+                        word &= word - 1;
+                        idx[sub][pos++] = (uint32_t)(lo + b);
+                    }
+                }
             }
         }
+        __syncthreads();
+        if (active) {
+            const int m = cnt[sub];
+            for (int k = lane; k < m; k += LANES)
+                g1_madd(acc, acc, table[idx[sub][k]]);
+        }
     }
+    __syncthreads();
     red[sub][lane] = acc;
     __syncthreads();
-    for (int s = MASK_LANES / 2; s > 0; s >>= 1) {
+    for (int s = LANES / 2; s > 0; s >>= 1) {
         if (lane < s) {
             g1_t t;
             g1_add(t, red[sub][lane], red[sub][lane + s]);
@@ -1323,6 +1335,21 @@ k_mask_aggregate(const g1aff_t *table, int n, const uint8_t *bitmaps,
             out[item] = red[sub][0];
         }
     }
+}
+/* LANES dispatch: a full wave per item once the per-item scan is long
+ * enough to keep it busy (n >= 16384 also covers config4's 65536). */
+static inline void launch_mask_aggregate(const g1aff_t *table, int n,
+        const uint8_t *bm, int bm_stride, const g1_t *full_sum,
+        g1_t *out, int batch) {
+    int blocks = (batch + MASK_SUBS - 1) / MASK_SUBS;
+    if (n >= 16384)
+        hipLaunchKernelGGL((k_mask_aggregate<64>), dim3(blocks),
+                           dim3(64 * MASK_SUBS), 0, 0,
+                           table, n, bm, bm_stride, full_sum, out, batch);
+    else
+        hipLaunchKernelGGL((k_mask_aggregate<MASK_LANES>), dim3(blocks),
+                           dim3(MASK_BLOCK), 0, 0,
+                           table, n, bm, bm_stride, full_sum, out, batch);
 }
 
 __global__ void k_hash_to_g2(const uint8_t *msgs, int mlen, g2_t *out,
@@ -1728,9 +1755,8 @@ extern "C" hbls_committee_t *hbls_committee_build(const uint8_t *pks48, size_t n
             std::vector<uint8_t> host_ones(bm, 0);
             for (size_t i = 0; i < n; i++) host_ones[i >> 3] |= 1 << (i & 7);
             if (hipMemcpy(ones.p, host_ones.data(), bm, hipMemcpyHostToDevice) == hipSuccess) {
-                hipLaunchKernelGGL(k_mask_aggregate, dim3(1), dim3(MASK_BLOCK), 0, 0,
-                                   c->d_table, (int)n, ones.as<uint8_t>(), (int)bm,
-                                   (const g1_t *)nullptr, d_fs, 1);
+                launch_mask_aggregate(c->d_table, (int)n, ones.as<uint8_t>(),
+                                      (int)bm, (const g1_t *)nullptr, d_fs, 1);
                 if (hipDeviceSynchronize() == hipSuccess)
                     c->d_full_sum = d_fs;
             }
@@ -1757,9 +1783,8 @@ extern "C" int hbls_mask_aggregate_g1(const hbls_committee_t *c, const uint8_t *
     if (dbm.err || dout.err || dser.err) return HBLS_ERR;
     HIP_OK(hipMemcpy(dbm.p, bitmap, bm, hipMemcpyHostToDevice));
     Timer tm;
-    hipLaunchKernelGGL(k_mask_aggregate, dim3(1), dim3(MASK_BLOCK), 0, 0,
-                       c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
-                       c->d_full_sum, dout.as<g1_t>(), 1);
+    launch_mask_aggregate(c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
+                          c->d_full_sum, dout.as<g1_t>(), 1);
     hipLaunchKernelGGL(k_g1_serialize, dim3(1), dim3(1), 0, 0,
                        dout.as<g1_t>(), dser.as<uint8_t>(), 1);
     tm.stop_and_store();
@@ -1786,10 +1811,8 @@ extern "C" int hbls_batch_agg_verify(const hbls_committee_t *c, const uint8_t *b
     hipEvent_t ev[5];
     for (int i = 0; i < 5; i++) (void)hipEventCreate(&ev[i]);
     (void)hipEventRecord(ev[0], 0);
-    hipLaunchKernelGGL(k_mask_aggregate, dim3((int)((batch + MASK_SUBS - 1) / MASK_SUBS)),
-                       dim3(MASK_BLOCK), 0, 0,
-                       c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
-                       c->d_full_sum, dagg.as<g1_t>(), (int)batch);
+    launch_mask_aggregate(c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
+                                         c->d_full_sum, dagg.as<g1_t>(), (int)batch);
     (void)hipEventRecord(ev[1], 0);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
@@ -2450,10 +2473,8 @@ extern "C" int hbls_mask_partials(const hbls_committee_t *c, const uint8_t *bitm
     HIP_OK(hipMemcpy(dbm.p, bitmaps, batch * bm, hipMemcpyHostToDevice));
     Timer tm;
     int nb = (int)((batch + 63) / 64);
-    hipLaunchKernelGGL(k_mask_aggregate, dim3((int)((batch + MASK_SUBS - 1) / MASK_SUBS)),
-                       dim3(MASK_BLOCK), 0, 0,
-                       c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
-                       c->d_full_sum, dagg.as<g1_t>(), (int)batch);
+    launch_mask_aggregate(c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
+                                         c->d_full_sum, dagg.as<g1_t>(), (int)batch);
     hipLaunchKernelGGL(k_g1_serialize, dim3(nb), dim3(64), 0, 0,
                        dagg.as<g1_t>(), dser.as<uint8_t>(), (int)batch);
     tm.stop_and_store();
@@ -2487,10 +2508,8 @@ extern "C" int hbls_batch_agg_verify_partials(
     HIP_OK(hipMemcpy(dmsg.p, msgs, batch * msg_len, hipMemcpyHostToDevice));
     Timer tm;
     int nb = (int)((batch + 63) / 64);
-    hipLaunchKernelGGL(k_mask_aggregate, dim3((int)((batch + MASK_SUBS - 1) / MASK_SUBS)),
-                       dim3(MASK_BLOCK), 0, 0,
-                       c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
-                       c->d_full_sum, dagg.as<g1_t>(), (int)batch);
+    launch_mask_aggregate(c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
+                                         c->d_full_sum, dagg.as<g1_t>(), (int)batch);
     if (n_ext)
         hipLaunchKernelGGL(k_add_partials, dim3(nb), dim3(64), 0, 0,
                            dagg.as<g1_t>(), dext.as<uint8_t>(), (int)n_ext,
