@@ -417,32 +417,38 @@ DEVN void g1_add(g1_t &r, const g1_t &p, const g1_t &q) {
     fp_mul_inl(r.z, t, h);
 }
 /* mixed add: r = p + affine q (q never infinity — table keys) */
-DEVN void g1_madd(g1_t &r, const g1_t &p, const g1aff_t &q) {
-    if (g1_is_inf(p)) { r.x = q.x; r.y = q.y; fp_one(r.z); return; }
-    fp_t z1z1, u2, s2, h, rr, hh, hhh, v, t;
-    fp_sqr_inl(z1z1, p.z);
-    fp_mul_inl(u2, q.x, z1z1);
-    fp_mul_inl(s2, q.y, p.z);
-    fp_mul_inl(s2, s2, z1z1);
-    fp_sub(h, u2, p.x);
-    fp_sub(rr, s2, p.y);
-    if (fp_is_zero(h)) {
-        if (fp_is_zero(rr)) { g1_dbl(r, p); return; }
-        g1_set_inf(r); return;
-    }
-    fp_sqr_inl(hh, h);
-    fp_mul_inl(hhh, hh, h);
-    fp_mul_inl(v, p.x, hh);
-    fp_sqr_inl(t, rr);
-    fp_sub(t, t, hhh);
-    fp_sub(t, t, v);
-    fp_sub(r.x, t, v);
-    fp_sub(t, v, r.x);
-    fp_mul_inl(t, rr, t);
-    fp_mul_inl(v, p.y, hhh);
-    fp_sub(r.y, t, v);
-    fp_mul_inl(r.z, p.z, h);
-}
+#define G1_MADD_BODY(r, p, q) do { \
+    if (g1_is_inf(p)) { (r).x = (q).x; (r).y = (q).y; fp_one((r).z); break; } \
+    fp_t z1z1, u2, s2, h, rr, hh, hhh, v, t; \
+    fp_sqr_inl(z1z1, (p).z); \
+    fp_mul_inl(u2, (q).x, z1z1); \
+    fp_mul_inl(s2, (q).y, (p).z); \
+    fp_mul_inl(s2, s2, z1z1); \
+    fp_sub(h, u2, (p).x); \
+    fp_sub(rr, s2, (p).y); \
+    if (fp_is_zero(h)) { \
+        if (fp_is_zero(rr)) { g1_dbl(r, p); break; } \
+        g1_set_inf(r); break; \
+    } \
+    fp_sqr_inl(hh, h); \
+    fp_mul_inl(hhh, hh, h); \
+    fp_mul_inl(v, (p).x, hh); \
+    fp_sqr_inl(t, rr); \
+    fp_sub(t, t, hhh); \
+    fp_sub(t, t, v); \
+    fp_sub((r).x, t, v); \
+    fp_sub(t, v, (r).x); \
+    fp_mul_inl(t, rr, t); \
+    fp_mul_inl(v, (p).y, hhh); \
+    fp_sub((r).y, t, v); \
+    fp_mul_inl((r).z, (p).z, h); \
+} while (0)
+DEVN void g1_madd(g1_t &r, const g1_t &p, const g1aff_t &q) { G1_MADD_BODY(r, p, q); }
+/* inline clone for the mask kernel's accumulate loop: the DEVN call passes
+ * the accumulator by reference, costing a 144 B scratch round-trip per
+ * point addition; the kernel has exactly one call site so inlining
+ * duplicates nothing. */
+DEV void g1_madd_i(g1_t &r, const g1_t &p, const g1aff_t &q) { G1_MADD_BODY(r, p, q); }
 DEV void g1_neg(g1_t &r, const g1_t &p) { r.x = p.x; fp_neg(r.y, p.y); r.z = p.z; }
 DEVN void g1_to_affine(g1aff_t &r, const g1_t &p) {
     fp_t zi, zi2, zi3;
@@ -1311,7 +1317,7 @@ k_mask_aggregate(const g1aff_t *table, int n, const uint8_t *bitmaps,
         if (active) {
             const int m = cnt[sub];
             for (int k = lane; k < m; k += LANES)
-                g1_madd(acc, acc, table[idx[sub][k]]);
+                g1_madd_i(acc, acc, table[idx[sub][k]]);
         }
     }
     __syncthreads();
