@@ -1375,12 +1375,12 @@ __global__ void k_wtab_to_affine(const g1_t *wj, g1aff_t *wtab, uint8_t *winf,
 /* windowed masked sum: one g1_madd per nonzero minority byte.  At ~1/3
  * minority density ~96% of bytes are nonzero, so no compaction is needed —
  * lanes stride bytes directly with negligible divergence. */
-__global__ void __launch_bounds__(64 * MASK_SUBS)
+template <int LANES>
+__global__ void __launch_bounds__(LANES * MASK_SUBS)
 k_mask_aggregate_w(const g1aff_t *wtab, const uint8_t *winf, int n,
                    const uint8_t *bitmaps, int bm_stride,
                    const g1_t *full_sum, g1_t *out, int batch) {
-    const int LANES = 64;
-    __shared__ g1_t red[MASK_SUBS][64];
+    __shared__ g1_t red[MASK_SUBS][LANES];
     __shared__ int cnt[MASK_SUBS];
     const int sub = threadIdx.x / LANES;
     const int lane = threadIdx.x % LANES;
@@ -1442,9 +1442,16 @@ static inline void launch_mask_aggregate(const g1aff_t *table, int n,
         const g1aff_t *wtab = nullptr, const uint8_t *winf = nullptr) {
     int blocks = (batch + MASK_SUBS - 1) / MASK_SUBS;
     if (wtab != nullptr) {
-        hipLaunchKernelGGL(k_mask_aggregate_w, dim3(blocks),
-                           dim3(64 * MASK_SUBS), 0, 0,
-                           wtab, winf, n, bm, bm_stride, full_sum, out, batch);
+        /* small bitmaps leave 64 lanes with so few byte-adds that the
+         * reduction tree dominates: use 16 lanes below 2048 bytes */
+        if (bm_stride >= 2048)
+            hipLaunchKernelGGL((k_mask_aggregate_w<64>), dim3(blocks),
+                               dim3(64 * MASK_SUBS), 0, 0,
+                               wtab, winf, n, bm, bm_stride, full_sum, out, batch);
+        else
+            hipLaunchKernelGGL((k_mask_aggregate_w<16>), dim3(blocks),
+                               dim3(16 * MASK_SUBS), 0, 0,
+                               wtab, winf, n, bm, bm_stride, full_sum, out, batch);
         return;
     }
     if (n >= 16384)
@@ -1855,9 +1862,10 @@ extern "C" hbls_committee_t *hbls_committee_build(const uint8_t *pks48, size_t n
     c->d_full_sum = nullptr;
     c->d_wtab = nullptr;
     c->d_winf = nullptr;
-    /* 8-bit window table: 255 subset sums per 8-key group (~200 MB at
-     * n=65536; skipped silently if the transient allocations fail) */
-    if (n >= 16384) {
+    /* 8-bit window table: 255 subset sums per 8-key group (n x 3060 B:
+     * ~12.5 MB at n=4096, ~200 MB at n=65536; skipped silently if the
+     * transient allocations fail) */
+    if (n >= 2048) {
         size_t groups = (n + 7) / 8, m = groups * 255;
         g1_t *d_wj = nullptr;
         g1aff_t *d_wt = nullptr;
