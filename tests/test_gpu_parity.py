@@ -152,6 +152,45 @@ def test_mask_aggregate_parity(core, capi, n):
         assert gc.mask_aggregate(full) == oc.mask_aggregate(full)
 
 
+def test_mask_windowed_edges(core, capi):
+    """Windowed mask path (n >= 2048): ragged tail-byte masking and the
+    infinity-flag branch — a key and its negation in the same 8-key group
+    sum to the point at infinity, which the affine window table cannot
+    represent (hbls_dev.hip k_mask_aggregate_w / winf)."""
+    import random
+    n = 2053                       # >= 2048 gate, ragged final byte (5 bits)
+    rng = random.Random(99)
+    sks = b"".join(sk_bytes(i) for i in range(n))
+    pks = bytearray(core.batch_pk_from_sk(sks, n))
+    # keys 8 and 9 (same byte group): make 9 the negation of 8 by flipping
+    # the herumi y-parity bit (bit 7 of the last byte)
+    pks[9 * 48:10 * 48] = pks[8 * 48:9 * 48]
+    pks[10 * 48 - 1] ^= 0x80
+    pks = bytes(pks)
+    gc = core.Committee(pks, n)
+    oc = capi.Committee(pks, n)
+    nb = (n + 7) // 8
+    # exactly the negation pair: its group's window entry is infinity
+    bm = bytearray(nb)
+    bm[1] = 0x03                   # bits 8 and 9
+    cases = [bytes(bm), bytes(nb)]
+    # full mask (ragged tail: bits 2053..2055 must stay clear)
+    full = bytearray([0xFF] * nb)
+    full[-1] = (1 << (n % 8)) - 1
+    cases.append(bytes(full))
+    # negation pair + random sparse and dense masks
+    for density in (0.1, 0.9):
+        r = bytearray(nb)
+        for i in range(n):
+            if rng.random() < density:
+                r[i >> 3] |= 1 << (i & 7)
+        r[1] |= 0x03
+        r[-1] &= (1 << (n % 8)) - 1
+        cases.append(bytes(r))
+    for bm in cases:
+        assert gc.mask_aggregate(bm) == oc.mask_aggregate(bm)
+
+
 def test_agg_verify_parity(core, capi, keys16):
     sks, pks, n = keys16
     gc = core.Committee(pks, n)
