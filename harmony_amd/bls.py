@@ -170,6 +170,12 @@ class Mask:
             raise ValueError(
                 f"mismatching bitmap lengths expected {self.length()} got {len(mask)}")
         self.Bitmap = bytearray(mask)
+        # The reference's SetMask only flips per-key bits, so its Bitmap padding
+        # stays zero (mask.go:113-134); clear bits >= len(Publics) so mask()
+        # round-trips and raw-bitmap comparisons match.
+        tail = len(self.Publics) & 7
+        if tail and self.Bitmap:
+            self.Bitmap[-1] &= (1 << tail) - 1
         self._agg_cache = None
 
     def set_bit(self, i: int, enable: bool):

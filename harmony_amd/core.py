@@ -216,29 +216,46 @@ class Committee:
             raise ValueError("committee_build: invalid pubkey in table")
         self.n = n
 
+    def _bitmap_len(self) -> int:
+        return (self.n + 7) >> 3
+
+    def _check_bitmaps(self, bitmaps: bytes, batch: int, what: str):
+        """Mask.SetMask errors on a length mismatch (crypto/bls/mask.go:113-118);
+        the FFI memcpys ceil(n/8) bytes per item, so a short buffer would read
+        out of bounds.  Validate before crossing the ABI."""
+        want = self._bitmap_len() * batch
+        if len(bitmaps) != want:
+            raise ValueError(
+                f"{what}: mismatching bitmap lengths expected {want} got {len(bitmaps)}")
+
     def mask_aggregate(self, bitmap: bytes) -> bytes:
+        self._check_bitmaps(bitmap, 1, "mask_aggregate")
         out = ctypes.create_string_buffer(48)
         _check(_lib.hbls_mask_aggregate_g1(self._h, bitmap, out), "mask_aggregate")
         return out.raw
 
     def agg_verify(self, bitmap: bytes, sig96: bytes, msg: bytes) -> bool:
+        self._check_bitmaps(bitmap, 1, "agg_verify")
         rc = _lib.hbls_agg_verify(self._h, bitmap, sig96, msg, len(msg))
         return _check(rc, "agg_verify") == HBLS_OK
 
     def batch_agg_verify(self, bitmaps: bytes, sigs: bytes, msgs: bytes,
                          mlen: int, batch: int):
+        self._check_bitmaps(bitmaps, batch, "batch_agg_verify")
         res = (ctypes.c_int32 * batch)()
         _check(_lib.hbls_batch_agg_verify(self._h, bitmaps, sigs, msgs, mlen,
                                           batch, res), "batch_agg_verify")
         return list(res)
 
     def mask_partials(self, bitmaps: bytes, batch: int) -> bytes:
+        self._check_bitmaps(bitmaps, batch, "mask_partials")
         out = ctypes.create_string_buffer(48 * batch)
         _check(_lib.hbls_mask_partials(self._h, bitmaps, batch, out), "mask_partials")
         return out.raw
 
     def batch_agg_verify_partials(self, bitmaps: bytes, ext48s: bytes, n_ext: int,
                                   sigs: bytes, msgs: bytes, mlen: int, batch: int):
+        self._check_bitmaps(bitmaps, batch, "batch_agg_verify_partials")
         res = (ctypes.c_int32 * batch)()
         _check(_lib.hbls_batch_agg_verify_partials(
             self._h, bitmaps, ext48s, n_ext, sigs, msgs, mlen, batch, res),

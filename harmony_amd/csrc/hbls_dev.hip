@@ -1635,6 +1635,15 @@ __global__ void k_add_partials(g1_t *aggs, const uint8_t *ext48s, int n_ext,
     ok[i] = 1;
 }
 
+/* config-4 epilogue: an item whose external partial failed to deserialize
+ * (ok[i]==0 from k_add_partials) must report bad input, not a verify result
+ * computed against the local-slice-only key sum. */
+__global__ void k_merge_pok(int32_t *results, const int32_t *pok, int batch) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= batch) return;
+    if (!pok[i]) results[i] = HBLS_ERR_BADINPUT;
+}
+
 /* sum of n serialized G2 signatures (AggregateSig batch form,
  * crypto/bls/mask.go:57-64): one 64-thread block, strided deserialize+add,
  * LDS tree.  flags: 1 ok, 0 bad input. */
@@ -2691,6 +2700,9 @@ extern "C" int hbls_batch_agg_verify_partials(
                            dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
     }
+    if (n_ext)
+        hipLaunchKernelGGL(k_merge_pok, dim3(nb), dim3(64), 0, 0,
+                           dres.as<int32_t>(), dpok.as<int32_t>(), (int)batch);
     tm.stop_and_store();
     HIP_OK(hipGetLastError());
     HIP_OK(hipMemcpy(results, dres.p, batch * 4, hipMemcpyDeviceToHost));

@@ -94,9 +94,17 @@ class Decider:
 
     def is_quorum_achieved_by_mask(self, bitmap: bytes) -> bool:
         """IsQuorumAchievedByMask (one-node-staked-vote.go:175-188 /
-        one-node-one-vote.go:75-86)"""
+        one-node-one-vote.go:75-86).  The reference builds the mask through
+        Mask.SetMask, which errors on a length mismatch (mask.go:113-118), and
+        its padding bits are always zero — enforce both here so an arbitrary
+        caller bitmap cannot inflate the count via padding/extra bytes."""
+        if len(bitmap) != (len(self.members) + 7) >> 3:
+            raise ValueError(
+                f"mismatching bitmap lengths expected {(len(self.members) + 7) >> 3} "
+                f"got {len(bitmap)}")
         if self.stakes is None:
-            n = sum(bin(b).count("1") for b in bitmap)
+            n = sum(1 for i in range(len(self.members))
+                    if bitmap[i >> 3] & (1 << (i & 7)))
             return n >= self.two_thirds_count()
         return self._mask_power(bitmap) > Fraction(2, 3)
 

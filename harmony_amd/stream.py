@@ -35,16 +35,29 @@ class StreamVerifier:
         self.window_checks = 0
         self._since_check = 0
 
-    def process_batch(self, key_idx, sigs_cat: bytes, blobs_cat: bytes, blob_len: int):
+    def process_batch(self, key_idx, sigs_cat: bytes, blobs_cat: bytes, blob_len: int,
+                      expected_digests: bytes = None):
         """One micro-batch of votes: blobs are hashed on-GPU (sender-auth
         digest path, checks.go:20-39 analog), signatures verified per sender
-        key in one launch, accepted votes folded into the aggregate."""
+        key in one launch, accepted votes folded into the aggregate.
+
+        expected_digests: batch*32 bytes of Keccak-256 digests the senders
+        committed to (the hash the message signature covers in
+        consensus_service.go:115-119); a vote whose blob hashes differently
+        is rejected before the signature check."""
         batch = len(key_idx)
         # keccak digests of the raw message blobs (crypto/hash/hash.go:9-15)
-        _digests = core.batch_keccak256(blobs_cat, blob_len, batch)
+        digests = core.batch_keccak256(blobs_cat, blob_len, batch)
+        digest_ok = [True] * batch
+        if expected_digests is not None:
+            if len(expected_digests) != 32 * batch:
+                raise ValueError("expected_digests must be batch*32 bytes")
+            digest_ok = [digests[32 * j:32 * (j + 1)] ==
+                         expected_digests[32 * j:32 * (j + 1)] for j in range(batch)]
         # per-vote verify of the commit payload signature
         msgs = self.payload * batch
         res = self.committee.batch_verify_votes(key_idx, sigs_cat, msgs, len(self.payload))
+        res = [r if digest_ok[j] else 0 for j, r in enumerate(res)]
         fresh = []
         for j, ok in enumerate(res):
             i = key_idx[j]

@@ -126,3 +126,50 @@ def test_vrf_primitive_api(env):
     proof = sks[2].sign_hash(digest)
     assert proof.verify_hash(ws[2].Object, digest)
     assert not proof.verify_hash(ws[2].Object, hashlib.sha256(b"other").digest())
+
+
+def test_vrf_evaluate_proof_to_hash(env):
+    """crypto/vrf/bls/bls_vrf.go:62-101 mirror: Evaluate
+    produces (beta, pi); ProofToHash accepts pi and returns beta; tampered
+    alpha or pi rejects; beta/pi match the oracle's CPU derivation."""
+    import hashlib
+    from harmony_amd import vrf
+    from oracle import capi
+    bls, sks, ws = env
+    signer = vrf.new_vrf_signer(sks[0])
+    verifier = vrf.new_vrf_verifier(ws[0].Object)
+    alpha = b"vrf-alpha-0"
+    beta, pi = signer.evaluate(alpha)
+    assert verifier.proof_to_hash(alpha, pi) == beta
+    # oracle parity: pi = SignHash(sk, sha256(alpha)), beta = sha256(pi)
+    pi_ref = capi.sign_hash(sks[0].serialize(), hashlib.sha256(alpha).digest())
+    assert pi == pi_ref
+    assert beta == hashlib.sha256(pi_ref).digest()
+    # wrong alpha rejects
+    with pytest.raises(vrf.ErrInvalidVRF):
+        verifier.proof_to_hash(b"vrf-alpha-1", pi)
+    # wrong verifier key rejects
+    with pytest.raises(vrf.ErrInvalidVRF):
+        vrf.new_vrf_verifier(ws[1].Object).proof_to_hash(alpha, pi)
+    # empty / undecodable proofs reject like the reference's ErrInvalidVRF
+    with pytest.raises(vrf.ErrInvalidVRF):
+        verifier.proof_to_hash(alpha, b"")
+    with pytest.raises(vrf.ErrInvalidVRF):
+        verifier.proof_to_hash(alpha, b"\xff" * 96)
+
+
+def test_committee_bitmap_length_validation(env):
+    """a short/long bitmap must raise before crossing the ABI
+    (the FFI memcpys ceil(n/8) bytes — OOB read otherwise)."""
+    from harmony_amd import core
+    bls, sks, ws = env
+    c = core.Committee(b"".join(w.Bytes for w in ws), len(ws))
+    msg = b"m" * 40
+    with pytest.raises(ValueError):
+        c.agg_verify(b"\xff\x00", b"\x00" * 96, msg)     # 8 keys -> 1 byte
+    with pytest.raises(ValueError):
+        c.mask_aggregate(b"")
+    with pytest.raises(ValueError):
+        c.batch_agg_verify(b"\xff", b"\x00" * 192, msg * 2, 40, 2)
+    with pytest.raises(ValueError):
+        c.mask_partials(b"\xff\xff\xff", 2)

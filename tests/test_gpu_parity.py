@@ -325,6 +325,12 @@ def test_config4_sharded_partials(core, capi):
     bad[0:48] = capi.pk_from_sk(sk_bytes(31))
     res2 = cB.batch_agg_verify_partials(bmsB, bytes(bad), 1, sigs, msgs, 48, batch)
     assert res2[0] == 0 and res2[1:] == [1, 1]
+    # MALFORMED partial (x >= p, not a point encoding) -> bad input, not a
+    # local-slice-only verify result (dpok must gate the result)
+    mal = bytearray(partialsA)
+    mal[0:48] = pr.fp_to_le48(pr.P)      # x == p: fp decode fails
+    res3 = cB.batch_agg_verify_partials(bmsB, bytes(mal), 1, sigs, msgs, 48, batch)
+    assert res3[0] == core.HBLS_ERR_BADINPUT and res3[1:] == [1, 1]
 
 
 def test_deserialize_fuzz_agreement(core, capi):

@@ -139,3 +139,36 @@ def test_abi_symbols_exported():
     assert len(names) >= 25
     missing = [n for n in set(names) if not hasattr(core._lib, n)]
     assert missing == []
+
+
+def test_mask_set_mask_clears_padding():
+    """the reference's SetMask only flips per-key bits so Bitmap
+    padding stays zero (mask.go:113-134) — set_mask must clear bits >= n."""
+    from harmony_amd.bls import Mask
+    ws = _wrappers(10)
+    m = Mask(ws)
+    m.set_mask(b"\xff\xff")          # padding bits 10..15 set by the caller
+    assert m.mask() == b"\xff\x03"   # cleared to the 10 real keys
+    assert m.count_enabled() == 10
+    m2 = Mask(_wrappers(16))         # n % 8 == 0: nothing to clear
+    m2.set_mask(b"\xff\xff")
+    assert m2.mask() == b"\xff\xff"
+
+
+def test_quorum_bitmap_length_and_padding():
+    """verify_seal path must reject mismatched bitmap lengths
+    (like Mask.SetMask, mask.go:113-118) and never count padding bits."""
+    from harmony_amd.quorum import Decider
+    d = Decider(_wrappers(10))
+    with pytest.raises(ValueError):
+        d.is_quorum_achieved_by_mask(b"\xff")            # short
+    with pytest.raises(ValueError):
+        d.is_quorum_achieved_by_mask(b"\xff\x03\x00")    # long
+    # 6 real votes + all 6 padding bits set: padding must NOT reach quorum (7)
+    assert not d.is_quorum_achieved_by_mask(b"\x3f\xfc")
+    # 7 real votes -> quorum
+    assert d.is_quorum_achieved_by_mask(b"\x7f\x00")
+    # staked policy validates length too
+    ds = Decider(_wrappers(4), stakes=[10, 10, 10, 70])
+    with pytest.raises(ValueError):
+        ds.is_quorum_achieved_by_mask(b"\x08\x00")

@@ -62,3 +62,30 @@ def test_stream_round(oracle_lib):
     # oracle cross-check of the final aggregate
     oc = oracle_lib.Committee(pks, n)
     assert oc.agg_verify(bytes(sv.bitmap), sv.agg_sig, payload) is True
+
+
+def test_stream_digest_check(oracle_lib):
+    """the sender-auth digest leg must be a real check — a vote
+    whose blob does not hash to the digest the sender committed to is
+    rejected before the signature check (checks.go:20-39 analog)."""
+    from harmony_amd import core
+    from harmony_amd.stream import StreamVerifier
+    n = 16
+    sks = [pr.fr_serialize(pr.synth_sk(i)) for i in range(n)]
+    pks = core.batch_pk_from_sk(b"".join(sks), n)
+    payload = pr.construct_commit_payload(5, pr.synth_msg(5), 1)
+    sv = StreamVerifier(pks, n, payload, window=10 ** 9)
+    blob_len = 256
+    chunk = list(range(n))
+    sigs = core.batch_sign(b"".join(sks), payload * n, len(payload), n)
+    blobs = b"".join((pr.synth_msg(i) * 10)[:blob_len] for i in chunk)
+    # expected digests from the oracle keccak; corrupt blob #4 after hashing
+    expected = b"".join(oracle_lib.keccak256(
+        blobs[blob_len * j:blob_len * (j + 1)]) for j in range(n))
+    tampered = bytearray(blobs)
+    tampered[blob_len * 4] ^= 0xFF
+    res = sv.process_batch(chunk, sigs, bytes(tampered), blob_len,
+                           expected_digests=expected)
+    assert res[4] == 0 and all(r == 1 for j, r in enumerate(res) if j != 4)
+    assert sv.accepted == n - 1 and sv.rejected == 1
+    assert sv.final_check() is True
