@@ -1635,6 +1635,22 @@ __global__ void k_add_partials(g1_t *aggs, const uint8_t *ext48s, int n_ext,
     ok[i] = 1;
 }
 
+/* sync-path seal-blob split (§8f-3): commitSigAndBitmap blobs
+ * (96B sig || ceil(n/8)B bitmap each, internal/chain/sig.go:22-35) are
+ * uploaded raw and parsed on-device — coalesced byte-granular grid-stride
+ * over the blob buffer, no host-side per-item copies. */
+__global__ void k_seal_split(const uint8_t *blobs, size_t blob_len, size_t bm,
+                             uint8_t *sigs96, uint8_t *bms, size_t batch) {
+    size_t total = batch * blob_len;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t t = (size_t)blockIdx.x * blockDim.x + threadIdx.x; t < total; t += stride) {
+        size_t j = t / blob_len, off = t - j * blob_len;
+        uint8_t v = blobs[t];
+        if (off < 96) sigs96[j * 96 + off] = v;
+        else bms[j * bm + (off - 96)] = v;
+    }
+}
+
 /* config-4 epilogue: an item whose external partial failed to deserialize
  * (ok[i]==0 from k_add_partials) must report bad input, not a verify result
  * computed against the local-slice-only key sum. */
@@ -1948,46 +1964,43 @@ extern "C" int hbls_mask_aggregate_g1(const hbls_committee_t *c, const uint8_t *
     return HBLS_OK;
 }
 
-extern "C" int hbls_batch_agg_verify(const hbls_committee_t *c, const uint8_t *bitmaps,
-                                     const uint8_t *sigs96, const uint8_t *msgs,
-                                     size_t msg_len, size_t batch, int32_t *results) {
-    int rc = require_gpu();
-    if (rc != HBLS_OK) return rc;
+/* the four-stage aggregate-verify pipeline over DEVICE-resident inputs
+ * (bitmaps, serialized sigs, messages) — shared by the host-buffer entry
+ * points and the HBM-direct seal-blob path. */
+static int run_agg_verify_pipeline(const hbls_committee_t *c, const uint8_t *d_bm,
+                                   const uint8_t *d_sig, const uint8_t *d_msg,
+                                   size_t msg_len, size_t batch, int32_t *results) {
     size_t bm = (c->n + 7) / 8;
-    DevBuf dbm(batch * bm), dsig(batch * 96), dmsg(batch * msg_len);
     DevBuf dagg(batch * sizeof(g1_t)), dhm(batch * sizeof(g2_t));
     DevBuf dsaff(batch * sizeof(g2aff_t)), dsflags(batch * 4), dhok(batch * 4), dres(batch * 4);
-    if (dbm.err || dsig.err || dmsg.err || dagg.err || dhm.err || dsaff.err ||
-        dsflags.err || dhok.err || dres.err) return HBLS_ERR;
-    HIP_OK(hipMemcpy(dbm.p, bitmaps, batch * bm, hipMemcpyHostToDevice));
-    HIP_OK(hipMemcpy(dsig.p, sigs96, batch * 96, hipMemcpyHostToDevice));
-    HIP_OK(hipMemcpy(dmsg.p, msgs, batch * msg_len, hipMemcpyHostToDevice));
+    if (dagg.err || dhm.err || dsaff.err || dsflags.err || dhok.err || dres.err)
+        return HBLS_ERR;
     int nb = (int)((batch + 63) / 64);
     hipEvent_t ev[5];
     for (int i = 0; i < 5; i++) (void)hipEventCreate(&ev[i]);
     (void)hipEventRecord(ev[0], 0);
-    launch_mask_aggregate(c->d_table, (int)c->n, dbm.as<uint8_t>(), (int)bm,
+    launch_mask_aggregate(c->d_table, (int)c->n, d_bm, (int)bm,
                                          c->d_full_sum, dagg.as<g1_t>(), (int)batch,
                                          c->d_wtab, c->d_winf);
     (void)hipEventRecord(ev[1], 0);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
         hipLaunchKernelGGL(k_hash_to_g2_coop, dim3(nbc), dim3(64), 0, 0,
-                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+                           d_msg, (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
                            (int)batch, g_fast_cofactor);
     } else {
         hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
-                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+                           d_msg, (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
                            (int)batch, g_fast_cofactor);
     }
     (void)hipEventRecord(ev[2], 0);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
         hipLaunchKernelGGL(k_g2_decompress_coop, dim3(nbc), dim3(64), 0, 0,
-                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+                           d_sig, dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     } else {
         hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
-                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+                           d_sig, dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     }
     (void)hipEventRecord(ev[3], 0);
     if ((int)batch <= coop_threshold()) {
@@ -2013,6 +2026,21 @@ extern "C" int hbls_batch_agg_verify(const hbls_committee_t *c, const uint8_t *b
     HIP_OK(hipGetLastError());
     HIP_OK(hipMemcpy(results, dres.p, batch * 4, hipMemcpyDeviceToHost));
     return HBLS_OK;
+}
+
+extern "C" int hbls_batch_agg_verify(const hbls_committee_t *c, const uint8_t *bitmaps,
+                                     const uint8_t *sigs96, const uint8_t *msgs,
+                                     size_t msg_len, size_t batch, int32_t *results) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    size_t bm = (c->n + 7) / 8;
+    DevBuf dbm(batch * bm), dsig(batch * 96), dmsg(batch * msg_len);
+    if (dbm.err || dsig.err || dmsg.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(dbm.p, bitmaps, batch * bm, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(dsig.p, sigs96, batch * 96, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(dmsg.p, msgs, batch * msg_len, hipMemcpyHostToDevice));
+    return run_agg_verify_pipeline(c, dbm.as<uint8_t>(), dsig.as<uint8_t>(),
+                                   dmsg.as<uint8_t>(), msg_len, batch, results);
 }
 
 extern "C" int hbls_agg_verify(const hbls_committee_t *c, const uint8_t *bitmap,
@@ -2712,21 +2740,30 @@ extern "C" int hbls_batch_agg_verify_partials(
 /* sync-path batch seal verification (stagedstreamsync/sig_verify.go:23-59,
  * legacysync/syncing.go:857): each item is a raw commitSigAndBitmap blob
  * (96B sig || ceil(n/8)B bitmap, internal/chain/sig.go:22-35) plus its
- * commit payload; parses on the host (byte split) and runs the batched
- * aggregate-verify pipeline. */
+ * commit payload; the raw blob window travels to HBM ONCE and is split
+ * on-device (k_seal_split) — no host-side per-item copies (§8f-3). */
 extern "C" int hbls_batch_seal_verify(const hbls_committee_t *c,
                                       const uint8_t *sig_bitmaps, size_t blob_len,
                                       const uint8_t *msgs, size_t msg_len,
                                       size_t batch, int32_t *results) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
     size_t bm = (c->n + 7) / 8;
     if (blob_len != 96 + bm) return HBLS_ERR_BADINPUT;
-    std::vector<uint8_t> sigs(batch * 96), bms(batch * bm);
-    for (size_t j = 0; j < batch; j++) {
-        memcpy(&sigs[j * 96], sig_bitmaps + j * blob_len, 96);
-        memcpy(&bms[j * bm], sig_bitmaps + j * blob_len + 96, bm);
-    }
-    return hbls_batch_agg_verify(c, bms.data(), sigs.data(), msgs, msg_len,
-                                 batch, results);
+    DevBuf dblob(batch * blob_len), dsig(batch * 96), dbm(batch * bm);
+    DevBuf dmsg(batch * msg_len);
+    if (dblob.err || dsig.err || dbm.err || dmsg.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(dblob.p, sig_bitmaps, batch * blob_len, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(dmsg.p, msgs, batch * msg_len, hipMemcpyHostToDevice));
+    size_t total = batch * blob_len;
+    int nb = (int)((total + 255) / 256);
+    if (nb > 4096) nb = 4096;
+    hipLaunchKernelGGL(k_seal_split, dim3(nb), dim3(256), 0, 0,
+                       dblob.as<uint8_t>(), blob_len, bm,
+                       dsig.as<uint8_t>(), dbm.as<uint8_t>(), batch);
+    HIP_OK(hipGetLastError());
+    return run_agg_verify_pipeline(c, dbm.as<uint8_t>(), dsig.as<uint8_t>(),
+                                   dmsg.as<uint8_t>(), msg_len, batch, results);
 }
 
 /* ConstructCommitPayload (consensus/signature/signature.go:12-24):
