@@ -79,6 +79,29 @@ def build_inputs(rank, batch):
     return sks, bitmaps, sk_sums, msgs
 
 
+def exchange_partials(partials: bytes, rank, world, batch, i0, share,
+                      dist, backend) -> bytes:
+    """config-4 exchange: all-gather every rank's batch x 48B serialized
+    masked partial sums, return the OTHER ranks' partials for my item share
+    (layout [ext][item], the hbls_batch_agg_verify_partials input).
+
+    The nccl(RCCL) and gloo branches run the identical byte path — the only
+    difference is the device the gather tensor lives on — which
+    tests/test_dist_gloo.py guards so an 8-GPU RCCL run needs no new code."""
+    import torch
+    t = torch.frombuffer(bytearray(partials), dtype=torch.uint8)
+    if backend == "nccl":
+        t = t.cuda()
+    outs = [torch.empty_like(t) for _ in range(world)]
+    dist.all_gather(outs, t)
+    gathered = [o.cpu().numpy().tobytes() for o in outs]
+    ext = b"".join(g for r, g in enumerate(gathered) if r != rank)
+    # slice my share of items out of each rank's partial block
+    return b"".join(e[48 * i0:48 * (i0 + share)]
+                    for e in [ext[k * batch * 48:(k + 1) * batch * 48]
+                              for k in range(world - 1)])
+
+
 def run_config4(args, rank, world, dist):
     """BASELINE configs[3]: one 65536-key committee, index-range sharded
     across ranks; per step every rank computes its slice's masked partial
@@ -130,17 +153,8 @@ def run_config4(args, rank, world, dist):
     def one_step(timed_check=True):
         partials = slice_table.mask_partials(slice_bms, batch)   # batch x 48
         if world > 1:
-            t = torch.frombuffer(bytearray(partials), dtype=torch.uint8)
-            if backend == "nccl":
-                t = t.cuda()
-            outs = [torch.empty_like(t) for _ in range(world)]
-            dist.all_gather(outs, t)
-            ext = b"".join(o.cpu().numpy().tobytes()
-                           for r, o in enumerate(outs) if r != rank)
-            # slice my share out of each rank's partial block
-            ext_my = b"".join(e[48 * i0:48 * (i0 + share)]
-                              for e in [ext[k * batch * 48:(k + 1) * batch * 48]
-                                        for k in range(world - 1)])
+            ext_my = exchange_partials(partials, rank, world, batch, i0, share,
+                                       dist, backend)
             n_ext = world - 1
         else:
             ext_my, n_ext = b"", 0

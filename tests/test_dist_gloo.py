@@ -117,3 +117,32 @@ def test_config4_partial_sum_exchange(oracle_lib):
 
 def test_config3_rank_sharded_replicas(oracle_lib):
     _run(_worker_config3, 29513)
+
+
+def _worker_exchange_guard(rank, world, port, q):
+    try:
+        _init(rank, world, port)
+        import bench
+        batch, share = 6, 3
+        i0 = rank * share
+        # deterministic fake partials: byte = f(rank, item, pos)
+        partials = bytes((rank * 101 + j * 7 + p) % 256
+                         for j in range(batch) for p in range(48))
+        ext = bench.exchange_partials(partials, rank, world, batch, i0, share,
+                                      dist, "gloo")
+        # expected: the OTHER rank's partials for MY items, same formula
+        other = (rank + 1) % world
+        want = bytes((other * 101 + j * 7 + p) % 256
+                     for j in range(i0, i0 + share) for p in range(48))
+        q.put((rank, ext == want))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"error: {e}"))
+
+
+def test_config4_exchange_backend_guard():
+    """the nccl(RCCL) and gloo branches of the config-4
+    exchange share one byte path (bench.exchange_partials) — this pins the
+    gathered layout/slicing, so an 8-GPU RCCL run exercises already-tested
+    byte logic (only the tensor device differs)."""
+    _run(_worker_exchange_guard, 29517)
