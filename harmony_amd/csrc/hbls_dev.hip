@@ -2574,6 +2574,78 @@ extern "C" long long hbls_fpmul_asm_check(void) {
 }
 
 /* measured fp_mul throughput (muls/s) for the given variant */
+/* occupancy-sweep twin of the fp_mul µbench: 64-thread blocks so the host
+ * can dial waves/SIMD exactly (de-risks the 1-wave register-file verify
+ * kernel design — how much dependent-mad latency is exposed with no
+ * co-resident waves when the stream is pure register VALU code?).
+ * chains: number of independent 4-limb... independent mul chains per lane
+ * (3 matches Karatsuba fp2_mul ILP, 1 is worst-case serial). */
+__global__ void __launch_bounds__(64) k_fpmul_bench_w(uint64_t *sink, int iters, int chains) {
+    fp_t a, b;
+#pragma unroll
+    for (int i = 0; i < 6; i++) {
+        a.l[i] = BLS_ONE_P[i] ^ (blockIdx.x * 64 + threadIdx.x);
+        b.l[i] = BLS_R2P[i] ^ (i * 1234567u);
+    }
+    a.l[5] &= 0x0fffffffffffffffULL;
+    b.l[5] &= 0x0fffffffffffffffULL;
+    uint32_t p32[12], a32[12], b32[12], x0[12], x1[12], x2[12], x3[12];
+#pragma unroll
+    for (int i = 0; i < 6; i++) {
+        p32[2 * i] = (uint32_t)BLS_P[i];
+        p32[2 * i + 1] = (uint32_t)(BLS_P[i] >> 32);
+        a32[2 * i] = (uint32_t)a.l[i];
+        a32[2 * i + 1] = (uint32_t)(a.l[i] >> 32);
+        b32[2 * i] = (uint32_t)b.l[i];
+        b32[2 * i + 1] = (uint32_t)(b.l[i] >> 32);
+    }
+    uint32_t pinv32 = (uint32_t)BLS_P_INV;
+#pragma unroll
+    for (int i = 0; i < 12; i++) { x0[i] = a32[i]; x1[i] = b32[i]; x2[i] = a32[i]; x3[i] = b32[i]; }
+    if (chains == 4) {
+        for (int it = 0; it < iters; it++) {
+            fp_mul32(x0, x0, a32, p32, pinv32);
+            fp_mul32(x1, x1, b32, p32, pinv32);
+            fp_mul32(x2, x2, a32, p32, pinv32);
+            fp_mul32(x3, x3, b32, p32, pinv32);
+        }
+    } else if (chains == 3) {
+        for (int it = 0; it < iters; it++) {
+            fp_mul32(x0, x0, a32, p32, pinv32);
+            fp_mul32(x1, x1, b32, p32, pinv32);
+            fp_mul32(x2, x2, a32, p32, pinv32);
+        }
+    } else {
+        for (int it = 0; it < iters; it++)
+            fp_mul32(x0, x0, a32, p32, pinv32);
+    }
+    if (x0[0] == 0xdeadbeef) sink[threadIdx.x] = x0[0] + x1[1] + x2[2] + x3[3];
+}
+
+/* rate in mul/s at a chosen occupancy: blocks64 64-thread blocks in flight.
+ * blocks64 = 1024 -> 1 wave/SIMD chip-wide, 8192 -> 8 waves/SIMD. */
+extern "C" double hbls_fpmul_bench_waves(int blocks64, int chains) {
+    if (require_gpu() != HBLS_OK) return 0.0;
+    DevBuf sink(64 * 8);
+    if (sink.err) return 0.0;
+    int iters = 4000;
+    int ch = (chains >= 4) ? 4 : (chains >= 3 ? 3 : 1);
+    hipLaunchKernelGGL(k_fpmul_bench_w, dim3(blocks64), dim3(64), 0, 0,
+                       sink.as<uint64_t>(), 200, ch);
+    (void)hipDeviceSynchronize();
+    hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0); (void)hipEventCreate(&e1);
+    (void)hipEventRecord(e0, 0);
+    hipLaunchKernelGGL(k_fpmul_bench_w, dim3(blocks64), dim3(64), 0, 0,
+                       sink.as<uint64_t>(), iters, ch);
+    (void)hipEventRecord(e1, 0);
+    if (hipEventSynchronize(e1) != hipSuccess) return 0.0;
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    (void)hipEventDestroy(e0); (void)hipEventDestroy(e1);
+    return (double)blocks64 * 64.0 * iters * (double)ch / (ms * 1e-3);
+}
+
 extern "C" double hbls_fpmul_bench_ops(int variant) {
     if (require_gpu() != HBLS_OK) return 0.0;
     DevBuf sink(256 * 8);
