@@ -62,6 +62,24 @@ def _load():
         ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t,
         ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_size_t,
         ctypes.POINTER(ctypes.c_int32)]
+    lib.hbls_stream_create.restype = ctypes.c_void_p
+    lib.hbls_stream_create.argtypes = [ctypes.c_void_p, ctypes.c_int]
+    lib.hbls_stream_free.argtypes = [ctypes.c_void_p]
+    lib.hbls_stream_set_rounds.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32), ctypes.c_int,
+        ctypes.c_char_p, ctypes.c_size_t]
+    lib.hbls_stream_process.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32),
+        ctypes.POINTER(ctypes.c_uint32), ctypes.c_char_p,
+        ctypes.POINTER(ctypes.c_uint32), ctypes.c_int,
+        ctypes.c_size_t, ctypes.POINTER(ctypes.c_int32)]
+    lib.hbls_stream_check.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32), ctypes.c_int,
+        ctypes.POINTER(ctypes.c_int32)]
+    lib.hbls_stream_get.argtypes = [ctypes.c_void_p, ctypes.c_uint32,
+                                    ctypes.c_char_p, ctypes.c_char_p]
+    lib.hbls_fpmul_bench_waves.restype = ctypes.c_double
+    lib.hbls_fpmul_bench_waves.argtypes = [ctypes.c_int, ctypes.c_int]
     return lib
 
 
@@ -282,5 +300,64 @@ class Committee:
         try:
             if getattr(self, "_h", None):
                 _lib.hbls_committee_free(self._h)
+        except Exception:
+            pass
+
+
+class Stream:
+    """Device-resident FBFT vote stream (config 5): per-round hash points,
+    bitmaps and running G2 aggregates live in HBM; one hbls_stream_process
+    call per tick.  Mirrors the leader's per-message loop
+    (consensus/leader.go:221-309) — see include/hbls.h."""
+
+    def __init__(self, committee: "Committee", max_rounds: int):
+        self.committee = committee        # keep alive: C ctx borrows its table
+        self.n = committee.n
+        self.max_rounds = max_rounds
+        self._h = _lib.hbls_stream_create(committee._h, max_rounds)
+        if not self._h:
+            if device_count() == 0:
+                raise NoGpuError("stream_create: no AMD GPU")
+            raise HblsError("stream_create failed")
+
+    def set_rounds(self, slots, payloads_cat: bytes, payload_len: int):
+        k = len(slots)
+        if payload_len * k != len(payloads_cat):
+            raise ValueError("payloads must be k * payload_len bytes")
+        arr = (ctypes.c_uint32 * k)(*slots)
+        _check(_lib.hbls_stream_set_rounds(self._h, arr, k, payloads_cat,
+                                           payload_len), "stream_set_rounds")
+
+    def process(self, key_idx, round_idx, sigs_cat: bytes):
+        batch = len(key_idx)
+        if len(round_idx) != batch or len(sigs_cat) != 96 * batch:
+            raise ValueError("key_idx/round_idx/sigs length mismatch")
+        active = sorted(set(round_idx))
+        ki = (ctypes.c_uint32 * batch)(*key_idx)
+        ri = (ctypes.c_uint32 * batch)(*round_idx)
+        act = (ctypes.c_uint32 * len(active))(*active)
+        res = (ctypes.c_int32 * batch)()
+        _check(_lib.hbls_stream_process(self._h, ki, ri, sigs_cat, act,
+                                        len(active), batch, res), "stream_process")
+        return list(res)
+
+    def check(self, slots):
+        k = len(slots)
+        arr = (ctypes.c_uint32 * k)(*slots)
+        ok = (ctypes.c_int32 * k)()
+        _check(_lib.hbls_stream_check(self._h, arr, k, ok), "stream_check")
+        return [bool(v == 1) for v in ok]
+
+    def get(self, slot: int):
+        """returns (bitmap bytes, serialized aggregate sig 96B)"""
+        bm = ctypes.create_string_buffer((self.n + 7) // 8)
+        agg = ctypes.create_string_buffer(96)
+        _check(_lib.hbls_stream_get(self._h, slot, bm, agg), "stream_get")
+        return bm.raw, agg.raw
+
+    def __del__(self):
+        try:
+            if getattr(self, "_h", None):
+                _lib.hbls_stream_free(self._h)
         except Exception:
             pass

@@ -1497,14 +1497,18 @@ __global__ void k_verify(const g1_t *aggpubs, const g2_t *hms, const g2aff_t *si
     results[i] = verify_pairing(aggpubs[i], hms[i], sig_inf ? dummy : sigs[i], sig_inf);
 }
 
-/* per-signer vote verify: pub = table[key_idx[i]] */
+/* per-signer vote verify: pub = table[key_idx[i]]; hm_idx (optional) maps
+ * each vote to a shared per-round hash point (stream path: 16 rounds share
+ * 16 hash-to-G2 results across thousands of votes) — null means hms[i]. */
 __global__ void k_verify_votes(const g1aff_t *table, int n, const uint32_t *key_idx,
-                               const g2_t *hms, const g2aff_t *sigs,
+                               const g2_t *hms, const uint32_t *hm_idx,
+                               const g2aff_t *sigs,
                                const int32_t *sig_flags, const int32_t *hm_ok,
                                int32_t *results, int batch) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= batch) return;
-    if (!hm_ok[i] || sig_flags[i] == 0 || key_idx[i] >= (uint32_t)n) {
+    int h = hm_idx ? (int)hm_idx[i] : i;
+    if (!hm_ok[h] || sig_flags[i] == 0 || key_idx[i] >= (uint32_t)n) {
         results[i] = HBLS_ERR_BADINPUT; return;
     }
     g1_t pub;
@@ -1513,7 +1517,7 @@ __global__ void k_verify_votes(const g1aff_t *table, int n, const uint32_t *key_
     fp_one(pub.z);
     g2aff_t dummy;
     bool sig_inf = sig_flags[i] == 2;
-    results[i] = verify_pairing(pub, hms[i], sig_inf ? dummy : sigs[i], sig_inf);
+    results[i] = verify_pairing(pub, hms[h], sig_inf ? dummy : sigs[i], sig_inf);
 }
 
 __global__ void k_g2_serialize(const g2_t *pts, uint8_t *out96, const int32_t *ok, int batch) {
@@ -2088,11 +2092,13 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
         hipLaunchKernelGGL(k_verify_votes_coop, dim3(nbc), dim3(64), 0, 0,
                            c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
+                           (const uint32_t *)nullptr,
                            dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
                            dres.as<int32_t>(), (int)batch);
     } else {
         hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
                            c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
+                           (const uint32_t *)nullptr,
                            dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
                            dres.as<int32_t>(), (int)batch);
     }
@@ -2920,5 +2926,324 @@ extern "C" int hbls_batch_keccak256(const uint8_t *msgs, size_t msg_len, size_t 
     tm.stop_and_store();
     HIP_OK(hipGetLastError());
     HIP_OK(hipMemcpy(out32s, dout.p, batch * 32, hipMemcpyDeviceToHost));
+    return HBLS_OK;
+}
+
+/* ======================== device-resident vote stream ========================
+ * Config-5 streaming FBFT rounds (consensus/leader.go:221-309 per-message
+ * loop) as a DEVICE-RESIDENT context: the committee table, per-round
+ * hash-to-G2 points, participation bitmaps and running G2 aggregates all
+ * live in HBM; one tick = one upload (key/round indices + signatures), a
+ * fixed launch chain (decompress -> verify -> dedup -> per-round
+ * accumulate), one 4B/vote download.  This replaces the round-1 python
+ * shape (per-call copies + per-round k_g2_sum launches) that capped the
+ * stream at ~3.4k msgs/s. */
+
+struct hbls_stream {
+    const hbls_committee_t *c;
+    int max_rounds;
+    int nwords;           /* committee bitmap words (32-bit) per round */
+    g2_t *d_hm;           /* per-round hash point */
+    int32_t *d_hm_ok;
+    uint32_t *d_bitmap;   /* max_rounds x nwords, little-endian bit order */
+    g2_t *d_agg;          /* per-round aggregate signature (jacobian) */
+};
+
+__global__ void k_stream_reset(uint32_t *bitmap, g2_t *agg, const uint32_t *slots,
+                               int k, int nwords) {
+    int s = blockIdx.x;
+    if (s >= k) return;
+    uint32_t r = slots[s];
+    for (int w = threadIdx.x; w < nwords; w += blockDim.x)
+        bitmap[(size_t)r * nwords + w] = 0;
+    if (threadIdx.x == 0) g2_set_inf(agg[r]);
+}
+
+/* scatter freshly hashed round payloads into the per-round slots */
+__global__ void k_stream_scatter_hm(g2_t *hm, int32_t *hm_ok, const g2_t *src,
+                                    const int32_t *src_ok, const uint32_t *slots, int k) {
+    int s = blockIdx.x * blockDim.x + threadIdx.x;
+    if (s >= k) return;
+    hm[slots[s]] = src[s];
+    hm_ok[slots[s]] = src_ok[s];
+}
+
+/* round-index validation + clamp (an OOB round index must not fault the
+ * verify kernel's hm gather; flagged items report HBLS_ERR_BADINPUT) */
+__global__ void k_stream_clamp(const uint32_t *round_idx, int max_rounds,
+                               uint32_t *clamped, int32_t *okflag, int batch) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= batch) return;
+    uint32_t r = round_idx[i];
+    int ok = r < (uint32_t)max_rounds;
+    clamped[i] = ok ? r : 0;
+    okflag[i] = ok;
+}
+
+/* dedup: first valid vote per (round, key) wins the bitmap bit; later valid
+ * votes become result=2 (valid duplicate — the reference's quorum dedup,
+ * quorum.go:354-394; within one tick the winner is unordered, as the
+ * reference's concurrent pubsub validators are). */
+__global__ void k_stream_dedup(uint32_t *bitmap, int nwords, const uint32_t *key_idx,
+                               const uint32_t *round_idx, int32_t *results, int batch) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= batch) return;
+    if (results[i] != 1) return;
+    uint32_t k = key_idx[i], r = round_idx[i];
+    uint32_t bit = 1u << (k & 31);
+    uint32_t old = atomicOr(&bitmap[(size_t)r * nwords + (k >> 5)], bit);
+    if (old & bit) results[i] = 2;
+}
+
+/* per-round accumulate of accepted signatures: one 64-thread block per
+ * active round slot, strided G2 adds + LDS tree (k_g2_sum shape), folded
+ * into the resident aggregate. */
+__global__ void __launch_bounds__(64) k_stream_accum(
+        g2_t *agg, const uint32_t *slots, int kslots, const uint32_t *round_idx,
+        const g2aff_t *sigs, const int32_t *results, int batch) {
+    int s = blockIdx.x;
+    if (s >= kslots) return;
+    uint32_t r = slots[s];
+    __shared__ g2_t red[64];
+    g2_t acc;
+    g2_set_inf(acc);
+    for (int i = threadIdx.x; i < batch; i += 64) {
+        if (round_idx[i] == r && results[i] == 1) {
+            g2_t q;
+            g2_from_affine(q, sigs[i]);
+            g2_add(acc, acc, q);
+        }
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int st = 32; st > 0; st >>= 1) {
+        if (threadIdx.x < st) {
+            g2_t t;
+            g2_add(t, red[threadIdx.x], red[threadIdx.x + st]);
+            red[threadIdx.x] = t;
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        g2_t t;
+        g2_add(t, agg[r], red[0]);
+        agg[r] = t;
+    }
+}
+
+/* gather per-round check inputs: byte bitmaps (the word array IS the LE
+ * byte layout), hm points, and the aggregate as affine + inf flag */
+__global__ void k_stream_gather_check(const uint32_t *bitmap, int nwords, int bmbytes,
+                                      const g2_t *hm, const int32_t *hm_ok,
+                                      const g2_t *agg, const uint32_t *slots, int k,
+                                      uint8_t *bm_out, g2_t *hm_out, int32_t *hmok_out,
+                                      g2aff_t *sig_out, int32_t *sflags_out) {
+    int s = blockIdx.x;
+    if (s >= k) return;
+    uint32_t r = slots[s];
+    const uint8_t *src = (const uint8_t *)(bitmap + (size_t)r * nwords);
+    for (int j = threadIdx.x; j < bmbytes; j += blockDim.x)
+        bm_out[(size_t)s * bmbytes + j] = src[j];
+    if (threadIdx.x == 0) {
+        hm_out[s] = hm[r];
+        hmok_out[s] = hm_ok[r];
+        g2_t a = agg[r];
+        if (g2_is_inf(a)) {
+            sflags_out[s] = 2;
+        } else {
+            sflags_out[s] = 1;
+            g2_to_affine(sig_out[s], a);
+        }
+    }
+}
+
+extern "C" void hbls_stream_free(hbls_stream *s);
+
+extern "C" hbls_stream *hbls_stream_create(const hbls_committee_t *c, int max_rounds) {
+    if (require_gpu() != HBLS_OK || !c || max_rounds <= 0) return nullptr;
+    hbls_stream *s = new hbls_stream();
+    s->c = c;
+    s->max_rounds = max_rounds;
+    s->nwords = (int)((c->n + 31) / 32);
+    size_t R = (size_t)max_rounds;
+    if (hipMalloc(&s->d_hm, R * sizeof(g2_t)) != hipSuccess ||
+        hipMalloc(&s->d_hm_ok, R * 4) != hipSuccess ||
+        hipMalloc(&s->d_bitmap, R * s->nwords * 4) != hipSuccess ||
+        hipMalloc(&s->d_agg, R * sizeof(g2_t)) != hipSuccess) {
+        hbls_stream_free(s);
+        return nullptr;
+    }
+    (void)hipMemset(s->d_hm_ok, 0, R * 4);
+    (void)hipMemset(s->d_bitmap, 0, R * s->nwords * 4);
+    /* aggregates start at the identity */
+    std::vector<uint32_t> all(R);
+    for (size_t i = 0; i < R; i++) all[i] = (uint32_t)i;
+    DevBuf dslots(R * 4);
+    if (dslots.err) { hbls_stream_free(s); return nullptr; }
+    (void)hipMemcpy(dslots.p, all.data(), R * 4, hipMemcpyHostToDevice);
+    hipLaunchKernelGGL(k_stream_reset, dim3((uint32_t)R), dim3(64), 0, 0,
+                       s->d_bitmap, s->d_agg, dslots.as<uint32_t>(), (int)R, s->nwords);
+    if (hipDeviceSynchronize() != hipSuccess) { hbls_stream_free(s); return nullptr; }
+    return s;
+}
+
+extern "C" void hbls_stream_free(hbls_stream *s) {
+    if (!s) return;
+    if (s->d_hm) (void)hipFree(s->d_hm);
+    if (s->d_hm_ok) (void)hipFree(s->d_hm_ok);
+    if (s->d_bitmap) (void)hipFree(s->d_bitmap);
+    if (s->d_agg) (void)hipFree(s->d_agg);
+    delete s;
+}
+
+/* (re)open round slots: hash the payloads on device (one batched launch),
+ * clear bitmaps + aggregates.  payloads: k x plen bytes. */
+extern "C" int hbls_stream_set_rounds(hbls_stream *s, const uint32_t *slots, int k,
+                                      const uint8_t *payloads, size_t plen) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    if (k <= 0 || k > s->max_rounds) return HBLS_ERR_BADINPUT;
+    for (int i = 0; i < k; i++)
+        if (slots[i] >= (uint32_t)s->max_rounds) return HBLS_ERR_BADINPUT;
+    DevBuf dslots(k * 4), dpl((size_t)k * plen), dhm(k * sizeof(g2_t)), dok(k * 4);
+    if (dslots.err || dpl.err || dhm.err || dok.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(dslots.p, slots, k * 4, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(dpl.p, payloads, (size_t)k * plen, hipMemcpyHostToDevice));
+    Timer tm;
+    int nbc = (k + CV_ITEMS - 1) / CV_ITEMS;
+    hipLaunchKernelGGL(k_hash_to_g2_coop, dim3(nbc), dim3(64), 0, 0,
+                       dpl.as<uint8_t>(), (int)plen, dhm.as<g2_t>(), dok.as<int32_t>(),
+                       k, g_fast_cofactor);
+    hipLaunchKernelGGL(k_stream_scatter_hm, dim3((k + 63) / 64), dim3(64), 0, 0,
+                       s->d_hm, s->d_hm_ok, dhm.as<g2_t>(), dok.as<int32_t>(),
+                       dslots.as<uint32_t>(), k);
+    hipLaunchKernelGGL(k_stream_reset, dim3(k), dim3(64), 0, 0,
+                       s->d_bitmap, s->d_agg, dslots.as<uint32_t>(), k, s->nwords);
+    tm.stop_and_store();
+    HIP_OK(hipGetLastError());
+    return HBLS_OK;
+}
+
+/* one tick: verify a batch of votes against (key_idx, round_idx), dedup via
+ * the resident bitmaps, fold accepted signatures into the resident
+ * aggregates.  results per vote: 1 accepted, 2 valid duplicate, 0 invalid
+ * signature, HBLS_ERR_BADINPUT malformed.  active_slots: the distinct round
+ * slots present in this batch (hosts know; keeps the accumulate launch
+ * exact). */
+extern "C" int hbls_stream_process(hbls_stream *s, const uint32_t *key_idx,
+                                   const uint32_t *round_idx, const uint8_t *sigs96,
+                                   const uint32_t *active_slots, int n_active,
+                                   size_t batch, int32_t *results) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    if (batch == 0 || n_active <= 0) return HBLS_ERR_BADINPUT;
+    for (int i = 0; i < n_active; i++)
+        if (active_slots[i] >= (uint32_t)s->max_rounds) return HBLS_ERR_BADINPUT;
+    DevBuf didx(batch * 4), dridx(batch * 4), dsig(batch * 96);
+    DevBuf dclamp(batch * 4), dpok(batch * 4);
+    DevBuf dsaff(batch * sizeof(g2aff_t)), dsflags(batch * 4), dres(batch * 4);
+    DevBuf dact(n_active * 4);
+    if (didx.err || dridx.err || dsig.err || dclamp.err || dpok.err ||
+        dsaff.err || dsflags.err || dres.err || dact.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(didx.p, key_idx, batch * 4, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(dridx.p, round_idx, batch * 4, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(dsig.p, sigs96, batch * 96, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(dact.p, active_slots, n_active * 4, hipMemcpyHostToDevice));
+    Timer tm;
+    int nb = (int)((batch + 63) / 64);
+    hipLaunchKernelGGL(k_stream_clamp, dim3(nb), dim3(64), 0, 0,
+                       dridx.as<uint32_t>(), s->max_rounds,
+                       dclamp.as<uint32_t>(), dpok.as<int32_t>(), (int)batch);
+    if ((int)batch <= coop_threshold()) {
+        int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
+        hipLaunchKernelGGL(k_g2_decompress_coop, dim3(nbc), dim3(64), 0, 0,
+                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+        hipLaunchKernelGGL(k_verify_votes_coop, dim3(nbc), dim3(64), 0, 0,
+                           s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
+                           s->d_hm, dclamp.as<uint32_t>(),
+                           dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), s->d_hm_ok,
+                           dres.as<int32_t>(), (int)batch);
+    } else {
+        hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
+                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+        hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
+                           s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
+                           s->d_hm, dclamp.as<uint32_t>(),
+                           dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), s->d_hm_ok,
+                           dres.as<int32_t>(), (int)batch);
+    }
+    hipLaunchKernelGGL(k_merge_pok, dim3(nb), dim3(64), 0, 0,
+                       dres.as<int32_t>(), dpok.as<int32_t>(), (int)batch);
+    hipLaunchKernelGGL(k_stream_dedup, dim3(nb), dim3(64), 0, 0,
+                       s->d_bitmap, s->nwords, didx.as<uint32_t>(),
+                       dclamp.as<uint32_t>(), dres.as<int32_t>(), (int)batch);
+    hipLaunchKernelGGL(k_stream_accum, dim3(n_active), dim3(64), 0, 0,
+                       s->d_agg, dact.as<uint32_t>(), n_active, dclamp.as<uint32_t>(),
+                       dsaff.as<g2aff_t>(), dres.as<int32_t>(), (int)batch);
+    tm.stop_and_store();
+    HIP_OK(hipGetLastError());
+    HIP_OK(hipMemcpy(results, dres.p, batch * 4, hipMemcpyDeviceToHost));
+    return HBLS_OK;
+}
+
+/* pairing-check the given rounds' resident aggregates against their
+ * resident masks (validator.go:224-228 shape).  ok[i]: 1/0. */
+extern "C" int hbls_stream_check(hbls_stream *s, const uint32_t *slots, int k,
+                                 int32_t *ok) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    if (k <= 0) return HBLS_ERR_BADINPUT;
+    for (int i = 0; i < k; i++)
+        if (slots[i] >= (uint32_t)s->max_rounds) return HBLS_ERR_BADINPUT;
+    size_t bm = (s->c->n + 7) / 8;
+    DevBuf dslots(k * 4), dbm((size_t)k * bm), dhm(k * sizeof(g2_t)), dhok(k * 4);
+    DevBuf dsaff(k * sizeof(g2aff_t)), dsflags(k * 4);
+    DevBuf dagg(k * sizeof(g1_t)), dres(k * 4);
+    if (dslots.err || dbm.err || dhm.err || dhok.err || dsaff.err ||
+        dsflags.err || dagg.err || dres.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(dslots.p, slots, k * 4, hipMemcpyHostToDevice));
+    Timer tm;
+    hipLaunchKernelGGL(k_stream_gather_check, dim3(k), dim3(64), 0, 0,
+                       s->d_bitmap, s->nwords, (int)bm, s->d_hm, s->d_hm_ok,
+                       s->d_agg, dslots.as<uint32_t>(), k,
+                       dbm.as<uint8_t>(), dhm.as<g2_t>(), dhok.as<int32_t>(),
+                       dsaff.as<g2aff_t>(), dsflags.as<int32_t>());
+    launch_mask_aggregate(s->c->d_table, (int)s->c->n, dbm.as<uint8_t>(), (int)bm,
+                          s->c->d_full_sum, dagg.as<g1_t>(), k,
+                          s->c->d_wtab, s->c->d_winf);
+    if (k <= coop_threshold()) {
+        int nbc = (k + CV_ITEMS - 1) / CV_ITEMS;
+        hipLaunchKernelGGL(k_verify_coop, dim3(nbc), dim3(64), 0, 0,
+                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+                           dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), k);
+    } else {
+        hipLaunchKernelGGL(k_verify, dim3((k + 63) / 64), dim3(64), 0, 0,
+                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+                           dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), k);
+    }
+    tm.stop_and_store();
+    HIP_OK(hipGetLastError());
+    HIP_OK(hipMemcpy(ok, dres.p, k * 4, hipMemcpyDeviceToHost));
+    return HBLS_OK;
+}
+
+/* export one round's state: byte bitmap + serialized aggregate (the shape
+ * consensus_service.go:305-322 signs into the block) */
+extern "C" int hbls_stream_get(hbls_stream *s, uint32_t slot, uint8_t *bitmap_out,
+                               uint8_t agg96_out[96]) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    if (slot >= (uint32_t)s->max_rounds) return HBLS_ERR_BADINPUT;
+    size_t bm = (s->c->n + 7) / 8;
+    HIP_OK(hipMemcpy(bitmap_out, (uint8_t *)(s->d_bitmap + (size_t)slot * s->nwords),
+                     bm, hipMemcpyDeviceToHost));
+    DevBuf dser(96), dok(4);
+    if (dser.err || dok.err) return HBLS_ERR;
+    int32_t one = 1;
+    HIP_OK(hipMemcpy(dok.p, &one, 4, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_g2_serialize, dim3(1), dim3(64), 0, 0,
+                       s->d_agg + slot, dser.as<uint8_t>(), dok.as<int32_t>(), 1);
+    HIP_OK(hipGetLastError());
+    HIP_OK(hipMemcpy(agg96_out, dser.p, 96, hipMemcpyDeviceToHost));
     return HBLS_OK;
 }

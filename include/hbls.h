@@ -165,6 +165,35 @@ int hbls_batch_seal_verify(const hbls_committee_t *c,
 /* AggregateSig batch form (crypto/bls/mask.go:57-64): out = sum of n sigs */
 int hbls_g2_aggregate(const uint8_t *sigs96, size_t n, uint8_t out96[96]);
 
+/* ---- device-resident vote stream (config 5: the FBFT leader's per-message
+ * hot loop, consensus/leader.go:221-309, re-shaped for the GPU) ----
+ * A stream context holds, in HBM: per-round hash-to-G2 points
+ * (ConstructCommitPayload outputs), participation bitmaps
+ * (commitBitmap/Mask, crypto/bls/mask.go:226-242) and running aggregate
+ * signatures (AddNewVote/Sign.Add, quorum.go:354-394).  One tick uploads
+ * (key_idx, round_idx, sig) per vote and runs decompress -> verify ->
+ * dedup -> per-round accumulate entirely on device. */
+typedef struct hbls_stream hbls_stream_t;
+hbls_stream_t *hbls_stream_create(const hbls_committee_t *c, int max_rounds);
+void hbls_stream_free(hbls_stream_t *s);
+/* (re)open round slots with their commit payloads (hashed on device) */
+int hbls_stream_set_rounds(hbls_stream_t *s, const uint32_t *slots, int k,
+                           const uint8_t *payloads, size_t payload_len);
+/* one tick; results[j]: 1 accepted, 2 valid duplicate (quorum.go:354-394
+ * double-vote rejection), 0 invalid, <0 malformed.  active_slots lists the
+ * distinct round slots present in the batch. */
+int hbls_stream_process(hbls_stream_t *s, const uint32_t *key_idx,
+                        const uint32_t *round_idx, const uint8_t *sigs96,
+                        const uint32_t *active_slots, int n_active,
+                        size_t batch, int32_t *results);
+/* pairing-check rounds' resident aggregates vs their masks
+ * (validator.go:224-228); ok[i] = 1/0 */
+int hbls_stream_check(hbls_stream_t *s, const uint32_t *slots, int k, int32_t *ok);
+/* export a round's bitmap + serialized aggregate
+ * (consensus_service.go:305-322 commitSigAndBitmap shape) */
+int hbls_stream_get(hbls_stream_t *s, uint32_t slot, uint8_t *bitmap_out,
+                    uint8_t agg96_out[96]);
+
 /* batch Keccak-256 (consensus message digests, crypto/hash/hash.go:9-15) */
 int hbls_batch_keccak256(const uint8_t *msgs, size_t msg_len, size_t batch,
                          uint8_t *out32s);

@@ -89,3 +89,67 @@ def test_stream_digest_check(oracle_lib):
     assert res[4] == 0 and all(r == 1 for j, r in enumerate(res) if j != 4)
     assert sv.accepted == n - 1 and sv.rejected == 1
     assert sv.final_check() is True
+
+
+def test_device_stream_context(oracle_lib):
+    """core.Stream (hbls_stream_*): multi-round device-resident verify +
+    dedup + accumulate, cross-checked against the oracle round by round."""
+    from harmony_amd import core
+    n, R = 32, 3
+    sks = [pr.fr_serialize(pr.synth_sk(i)) for i in range(n)]
+    pks = core.batch_pk_from_sk(b"".join(sks), n)
+    committee = core.Committee(pks, n)
+    payloads = [pr.construct_commit_payload(r, pr.synth_msg(100 + r), r + 1)
+                for r in range(R)]
+    st = core.Stream(committee, R)
+    st.set_rounds(list(range(R)), b"".join(payloads), len(payloads[0]))
+
+    # votes: every key votes in round 0; half in round 1; round 2 stays empty.
+    # include: one invalid sig, one duplicate in-batch, one duplicate
+    # across ticks.
+    votes = []   # (round, key)
+    for i in range(n):
+        votes.append((0, i))
+    for i in range(0, n, 2):
+        votes.append((1, i))
+    sig_of = {}
+    for r, i in set(votes):
+        sig_of[(r, i)] = oracle_lib.sign_hash(sks[i], payloads[r])
+    batch1 = votes[:20]
+    sigs1 = b"".join(sig_of[v] for v in batch1)
+    # corrupt vote #3's signature (valid point, wrong message)
+    bad = oracle_lib.sign_hash(sks[batch1[3][1]], payloads[2])
+    sigs1 = sigs1[:96 * 3] + bad + sigs1[96 * 4:]
+    res1 = st.process([v[1] for v in batch1], [v[0] for v in batch1], sigs1)
+    assert res1[3] == 0 and all(r == 1 for j, r in enumerate(res1) if j != 3)
+
+    batch2 = votes[20:] + [batch1[5], votes[21]]     # cross-tick + in-batch dup
+    sigs2 = b"".join(sig_of[v] for v in batch2)
+    res2 = st.process([v[1] for v in batch2], [v[0] for v in batch2], sigs2)
+    n2 = len(batch2)
+    assert res2[n2 - 2] == 2                          # cross-tick duplicate
+    assert sorted([res2[1], res2[n2 - 1]]) == [1, 2]  # in-batch dup: one wins
+    assert all(r in (1, 2) for r in res2)
+
+    # round checks: 0 and 1 verify; 2 is empty (identity vs empty mask = true
+    # herumi edge); after re-opening round 2 with a payload it stays true
+    assert st.check([0, 1, 2]) == [True, True, True]
+
+    # oracle cross-check of each round's exported bitmap + aggregate
+    oc = oracle_lib.Committee(pks, n)
+    for r in range(2):
+        bm, agg = st.get(r)
+        voters = set(i for (rr, i) in votes if rr == r)
+        assert bm == bytes(
+            sum(1 << (i & 7) for i in voters if i >> 3 == b) for b in range(4))
+        assert oc.agg_verify(bm, agg, payloads[r]) is True
+    # vote #3 (the corrupted one) retried with the right sig -> accepted now
+    k3 = batch1[3]
+    res3 = st.process([k3[1]], [k3[0]], sig_of[k3])
+    assert res3 == [1]
+    assert st.check([0]) == [True]
+    # unknown round slot raises
+    with pytest.raises(ValueError):
+        st.process([0], [R + 5], sig_of[(0, 0)])
+    with pytest.raises(ValueError):
+        st.check([R])
