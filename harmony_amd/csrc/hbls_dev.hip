@@ -1585,42 +1585,6 @@ __global__ void k_g2_subgroup_methods(const uint8_t *p96, int32_t *out2) {
 
 /* MSM v1: blocks of 256 threads; thread t of block b handles point b*256+t,
  * does full scalar mult, LDS tree reduce; second kernel reduces block results. */
-__global__ void __launch_bounds__(256)
-k_msm_partial(const uint8_t *points48, const uint8_t *scalars32, int n,
-              g1_t *partials, int32_t *ok) {
-    __shared__ g1_t red[256];
-    int i = blockIdx.x * 256 + threadIdx.x;
-    g1_t acc;
-    g1_set_inf(acc);
-    if (i < n) {
-        g1_t p;
-        uint64_t k[4];
-        if (!g1_deserialize(p, points48 + (size_t)i * 48, true) ||
-            !fr_from_le32(k, scalars32 + (size_t)i * 32)) {
-            atomicExch(ok, 0);
-        } else {
-            g1_mul(acc, p, k, 4);
-        }
-    }
-    red[threadIdx.x] = acc;
-    __syncthreads();
-    for (int s = 128; s > 0; s >>= 1) {
-        if (threadIdx.x < s) {
-            g1_t t;
-            g1_add(t, red[threadIdx.x], red[threadIdx.x + s]);
-            red[threadIdx.x] = t;
-        }
-        __syncthreads();
-    }
-    if (threadIdx.x == 0) partials[blockIdx.x] = red[0];
-}
-__global__ void k_g1_reduce(const g1_t *partials, int n, uint8_t *out48) {
-    g1_t acc;
-    g1_set_inf(acc);
-    for (int i = 0; i < n; i++) g1_add(acc, acc, partials[i]);
-    g1_serialize(out48, acc);
-}
-
 /* config-4 support: add n_ext serialized G1 partials (from other ranks'
  * committee slices) into each item's aggregate before the pairing check.
  * Partials are internal products (no subgroup check; identity = 48 zeros). */
@@ -2208,23 +2172,117 @@ extern "C" int hbls_g2_check(const uint8_t p96[96]) {
     return ok ? HBLS_OK : HBLS_FALSE;
 }
 
+/* ---- Pippenger bucket MSM (the `north_star`'s general-scalar MSM) ----
+ * radix-256 digits (c = 8): 32 windows x 255 buckets.  GPU shape:
+ *   prep     — thread/point: decompress + subgroup-check + digit transpose
+ *   buckets  — one WAVE per (window, bucket): lanes stride the digit row
+ *              (coalesced byte reads), accumulate matching points (mixed
+ *              adds; identity fast paths make empty lanes cheap), LDS tree
+ *   wreduce  — one thread per window: suffix sums S_w = sum b * B_b
+ *   combine  — Horner over windows (248 doublings + 31 adds), one thread
+ * Work ~ 32n mixed adds + 8160-wave trees vs n * (255 dbl + ~128 add) for
+ * the round-1 per-thread double-and-add. */
+__global__ void k_msm_prep(const uint8_t *points48, const uint8_t *scalars32, int n,
+                           g1aff_t *pts, uint8_t *digits /* 32 x n */, int32_t *ok) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    g1_t p;
+    uint64_t k[4];
+    if (!g1_deserialize(p, points48 + (size_t)i * 48, true) ||
+        !fr_from_le32(k, scalars32 + (size_t)i * 32)) {
+        atomicExch(ok, 0);
+        return;
+    }
+    /* serialized points are affine (z=1) after deserialize */
+    g1aff_t a;
+    a.x = p.x;
+    a.y = p.y;
+    bool inf = g1_is_inf(p);
+    pts[i] = a;
+    for (int w = 0; w < 32; w++) {
+        uint8_t d = (uint8_t)(k[w >> 3] >> (8 * (w & 7)));
+        /* infinity contributes nothing: digit 0 is never bucketed */
+        digits[(size_t)w * n + i] = inf ? 0 : d;
+    }
+}
+
+__global__ void __launch_bounds__(64) k_msm_buckets(
+        const g1aff_t *pts, const uint8_t *digits, int n, g1_t *buckets /* 32*255 */) {
+    int w = blockIdx.x / 255;
+    int b = blockIdx.x % 255 + 1;
+    const uint8_t *row = digits + (size_t)w * n;
+    g1_t acc;
+    g1_set_inf(acc);
+    for (int i = threadIdx.x; i < n; i += 64)
+        if (row[i] == b) g1_madd_i(acc, acc, pts[i]);
+    __shared__ g1_t red[64];
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int s = 32; s > 0; s >>= 1) {
+        if (threadIdx.x < s) {
+            g1_t t;
+            g1_add(t, red[threadIdx.x], red[threadIdx.x + s]);
+            red[threadIdx.x] = t;
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) buckets[w * 255 + b - 1] = red[0];
+}
+
+__global__ void __launch_bounds__(64) k_msm_wreduce(const g1_t *buckets, g1_t *wsums) {
+    int w = blockIdx.x * blockDim.x + threadIdx.x;
+    if (w >= 32) return;
+    g1_t running, sum;
+    g1_set_inf(running);
+    g1_set_inf(sum);
+    for (int b = 254; b >= 0; b--) {
+        g1_t t;
+        g1_add(t, running, buckets[w * 255 + b]);
+        running = t;
+        g1_add(t, sum, running);
+        sum = t;
+    }
+    wsums[w] = sum;
+}
+
+__global__ void k_msm_combine(const g1_t *wsums, uint8_t *out48) {
+    g1_t acc = wsums[31];
+    for (int w = 30; w >= 0; w--) {
+        for (int d = 0; d < 8; d++) {
+            g1_t t;
+            g1_dbl(t, acc);
+            acc = t;
+        }
+        g1_t t;
+        g1_add(t, acc, wsums[w]);
+        acc = t;
+    }
+    g1_serialize(out48, acc);
+}
+
 extern "C" int hbls_msm_g1(const uint8_t *points48, const uint8_t *scalars32, size_t n,
                            uint8_t out48[48]) {
     int rc = require_gpu();
     if (rc != HBLS_OK) return rc;
-    int nblocks = (int)((n + 255) / 256);
-    DevBuf dp(n * 48), ds(n * 32), dpart(nblocks * sizeof(g1_t)), dok(4), dout(48);
-    if (dp.err || ds.err || dpart.err || dok.err || dout.err) return HBLS_ERR;
+    if (n == 0) return HBLS_ERR_BADINPUT;
+    DevBuf dp(n * 48), ds(n * 32), dpts(n * sizeof(g1aff_t)), ddig(n * 32);
+    DevBuf dbuck(32 * 255 * sizeof(g1_t)), dws(32 * sizeof(g1_t)), dok(4), dout(48);
+    if (dp.err || ds.err || dpts.err || ddig.err || dbuck.err || dws.err ||
+        dok.err || dout.err) return HBLS_ERR;
     HIP_OK(hipMemcpy(dp.p, points48, n * 48, hipMemcpyHostToDevice));
     HIP_OK(hipMemcpy(ds.p, scalars32, n * 32, hipMemcpyHostToDevice));
     int32_t one = 1;
     HIP_OK(hipMemcpy(dok.p, &one, 4, hipMemcpyHostToDevice));
     Timer tm;
-    hipLaunchKernelGGL(k_msm_partial, dim3(nblocks), dim3(256), 0, 0,
+    hipLaunchKernelGGL(k_msm_prep, dim3((uint32_t)((n + 63) / 64)), dim3(64), 0, 0,
                        dp.as<uint8_t>(), ds.as<uint8_t>(), (int)n,
-                       dpart.as<g1_t>(), dok.as<int32_t>());
-    hipLaunchKernelGGL(k_g1_reduce, dim3(1), dim3(1), 0, 0,
-                       dpart.as<g1_t>(), nblocks, dout.as<uint8_t>());
+                       dpts.as<g1aff_t>(), ddig.as<uint8_t>(), dok.as<int32_t>());
+    hipLaunchKernelGGL(k_msm_buckets, dim3(32 * 255), dim3(64), 0, 0,
+                       dpts.as<g1aff_t>(), ddig.as<uint8_t>(), (int)n, dbuck.as<g1_t>());
+    hipLaunchKernelGGL(k_msm_wreduce, dim3(1), dim3(64), 0, 0,
+                       dbuck.as<g1_t>(), dws.as<g1_t>());
+    hipLaunchKernelGGL(k_msm_combine, dim3(1), dim3(1), 0, 0,
+                       dws.as<g1_t>(), dout.as<uint8_t>());
     tm.stop_and_store();
     HIP_OK(hipGetLastError());
     int32_t ok;
@@ -2621,6 +2679,26 @@ __global__ void __launch_bounds__(64) k_fpmul_bench_w(uint64_t *sink, int iters,
             fp_mul32(x1, x1, b32, p32, pinv32);
             fp_mul32(x2, x2, a32, p32, pinv32);
         }
+    } else if (chains == 5) {
+        /* hand-allocated asm CIOS body inlined, 3 chains (the register-file
+         * kernel's candidate multiplier; pinned regs serialize across
+         * chains — measures the 1-wave issue-rate story) */
+        fp_t y0 = a, y1 = b, y2 = a;
+        for (int it = 0; it < iters; it++) {
+            FP_MUL_ASM_BODY(y0, y0, a);
+            FP_MUL_ASM_BODY(y1, y1, b);
+            FP_MUL_ASM_BODY(y2, y2, a);
+        }
+        if (y0.l[0] == 0xdeadbeef) sink[threadIdx.x] = y0.l[0] + y1.l[1] + y2.l[2];
+    } else if (chains == 6) {
+        /* asm CIOS in called form, 3 chains */
+        fp_t y0 = a, y1 = b, y2 = a;
+        for (int it = 0; it < iters; it++) {
+            fp_mul_asm(y0, y0, a);
+            fp_mul_asm(y1, y1, b);
+            fp_mul_asm(y2, y2, a);
+        }
+        if (y0.l[0] == 0xdeadbeef) sink[threadIdx.x] = y0.l[0] + y1.l[1] + y2.l[2];
     } else {
         for (int it = 0; it < iters; it++)
             fp_mul32(x0, x0, a32, p32, pinv32);
@@ -2635,7 +2713,8 @@ extern "C" double hbls_fpmul_bench_waves(int blocks64, int chains) {
     DevBuf sink(64 * 8);
     if (sink.err) return 0.0;
     int iters = 4000;
-    int ch = (chains >= 4) ? 4 : (chains >= 3 ? 3 : 1);
+    int ch = chains;
+    if (chains < 3) ch = 1; else if (chains == 4) ch = 4; else if (chains > 6) ch = 3;
     hipLaunchKernelGGL(k_fpmul_bench_w, dim3(blocks64), dim3(64), 0, 0,
                        sink.as<uint64_t>(), 200, ch);
     (void)hipDeviceSynchronize();
@@ -2649,7 +2728,8 @@ extern "C" double hbls_fpmul_bench_waves(int blocks64, int chains) {
     float ms = 0;
     (void)hipEventElapsedTime(&ms, e0, e1);
     (void)hipEventDestroy(e0); (void)hipEventDestroy(e1);
-    return (double)blocks64 * 64.0 * iters * (double)ch / (ms * 1e-3);
+    double muls_per_iter = (ch == 4) ? 4.0 : (ch == 1 ? 1.0 : 3.0);
+    return (double)blocks64 * 64.0 * iters * muls_per_iter / (ms * 1e-3);
 }
 
 extern "C" double hbls_fpmul_bench_ops(int variant) {

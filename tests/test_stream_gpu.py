@@ -135,6 +135,11 @@ def test_device_stream_context(oracle_lib):
     # herumi edge); after re-opening round 2 with a payload it stays true
     assert st.check([0, 1, 2]) == [True, True, True]
 
+    # vote #3 (the corrupted one) retried with the right sig -> accepted now
+    k3 = batch1[3]
+    res3 = st.process([k3[1]], [k3[0]], sig_of[k3])
+    assert res3 == [1]
+    assert st.check([0]) == [True]
     # oracle cross-check of each round's exported bitmap + aggregate
     oc = oracle_lib.Committee(pks, n)
     for r in range(2):
@@ -143,11 +148,6 @@ def test_device_stream_context(oracle_lib):
         assert bm == bytes(
             sum(1 << (i & 7) for i in voters if i >> 3 == b) for b in range(4))
         assert oc.agg_verify(bm, agg, payloads[r]) is True
-    # vote #3 (the corrupted one) retried with the right sig -> accepted now
-    k3 = batch1[3]
-    res3 = st.process([k3[1]], [k3[0]], sig_of[k3])
-    assert res3 == [1]
-    assert st.check([0]) == [True]
     # unknown round slot raises
     with pytest.raises(ValueError):
         st.process([0], [R + 5], sig_of[(0, 0)])
