@@ -487,3 +487,44 @@ def test_deserialize_fuzz_extended(core, capi):
     mism48 = [c.hex() for c in c48 if core.g1_check(c) != capi.g1_check(c)]
     mism96 = [c.hex() for c in c96 if core.g2_check(c) != capi.g2_check(c)]
     assert mism48 == [] and mism96 == []
+
+
+def test_msm_pippenger_4096(core, capi):
+    """Pippenger bucket MSM parity at n=4096 random scalars.
+    Points are pk_i = sk_i*G, so sum_i s_i*P_i = (sum_i s_i*sk_i mod r)*G —
+    an exact python-int expectation with one C scalar mult.  Includes
+    duplicate points with equal digits (doubling branch), a zero scalar,
+    the identity point, and near-r scalars."""
+    import random
+    rng = random.Random(1234)
+    n = 4096
+    sks = b"".join(sk_bytes(i) for i in range(n))
+    pks = core.batch_pk_from_sk(sks, n)
+    pts = [pks[48 * i:48 * (i + 1)] for i in range(n)]
+    scalars = [rng.randrange(pr.R) for _ in range(n)]
+    scalars[0] = 0                      # zero scalar contributes nothing
+    scalars[1] = pr.R - 1               # top digits exercised
+    pts[2] = pts[3]                     # duplicate point
+    scalars[2] = scalars[3]             # ... with identical digits (dbl path)
+    pts[4] = b"\x00" * 48               # identity point, nonzero scalar
+    sk_ints = [pr.synth_sk(i) for i in range(n)]
+    acc = 0
+    for i in range(n):
+        if pts[i] != b"\x00" * 48:
+            j = i
+            if i == 2:
+                j = 3               # pts[2] was replaced by pts[3]
+            acc = (acc + scalars[i] * sk_ints[j]) % pr.R
+    expected = capi.pk_from_sk(pr.fr_serialize(acc))
+    got = core.msm_g1(b"".join(pts), b"".join(pr.fr_serialize(s) for s in scalars), n)
+    assert got == expected
+    # malformed inputs reject
+    import pytest as _pytest
+    bad_pts = bytearray(b"".join(pts))
+    bad_pts[0:48] = pr.fp_to_le48(pr.P)
+    with _pytest.raises(ValueError):
+        core.msm_g1(bytes(bad_pts), b"".join(pr.fr_serialize(s) for s in scalars), n)
+    bad_sc = bytearray(b"".join(pr.fr_serialize(s) for s in scalars))
+    bad_sc[0:32] = b"\xff" * 32         # scalar >= r
+    with _pytest.raises(ValueError):
+        core.msm_g1(b"".join(pts), bytes(bad_sc), n)
