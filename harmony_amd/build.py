@@ -15,6 +15,18 @@ CMD = [
     "hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-w",
     "-shared", "-fPIC", SRC, "-o", OUT,
 ]
+# instrumented twin (fp_mul call counter; loaded explicitly by experiments)
+OUT_COUNT = os.path.join(DIR, "libhbls_count.so")
+CMD_COUNT = CMD[:-1] + [OUT_COUNT, "-DHBLS_COUNT_MULS"]
+
+
+def build_count(force=False):
+    if not force and os.path.exists(OUT_COUNT) and \
+            os.path.getmtime(OUT_COUNT) > os.path.getmtime(SRC):
+        return OUT_COUNT
+    print("+", " ".join(CMD_COUNT), flush=True)
+    subprocess.check_call(CMD_COUNT)
+    return OUT_COUNT
 
 
 def build(force=False):

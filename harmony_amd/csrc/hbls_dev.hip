@@ -121,7 +121,18 @@ DEV void fp_dbl(fp_t &r, const fp_t &x) { fp_add(r, x, x); }
  * The body is shared between the default CALL form (noinline: a fully
  * inlined build thrashes the instruction cache and ran pathologically
  * slow) and an inline clone used only inside fp2_mul/fp2_sqr. */
+/* optional per-call fp_mul counter (perf-instrumentation builds only:
+ * -DHBLS_COUNT_MULS; settles the roofline's algorithmic-mul denominator
+ * with the GPU path's OWN count instead of the oracle's) */
+#ifdef HBLS_COUNT_MULS
+__device__ unsigned long long g_fp_mul_count;
+#define MULCOUNT() atomicAdd(&g_fp_mul_count, 1ull)
+#else
+#define MULCOUNT()
+#endif
+
 #define FP_MUL_BODY(r, x, y) do { \
+    MULCOUNT(); \
     uint32_t a_[12], b_[12], t_[13]; \
     _Pragma("unroll") \
     for (int i_ = 0; i_ < 6; i_++) { \
@@ -171,6 +182,7 @@ DEV void fp_dbl(fp_t &r, const fp_t &x) { fp_add(r, x, x); }
  * killing the pair-shuffle movs). */
 #include "fp_mul_asm.inc"
 #define FP_MUL_ASM_BODY(r, x, y) do { \
+    MULCOUNT(); \
     uint32_t a_[12], b_[12], t_[13]; \
     _Pragma("unroll") \
     for (int i_ = 0; i_ < 6; i_++) { \
@@ -3326,4 +3338,22 @@ extern "C" int hbls_stream_get(hbls_stream *s, uint32_t slot, uint8_t *bitmap_ou
     HIP_OK(hipGetLastError());
     HIP_OK(hipMemcpy(agg96_out, dser.p, 96, hipMemcpyDeviceToHost));
     return HBLS_OK;
+}
+
+extern "C" int hbls_mulcount_reset(void) {
+#ifdef HBLS_COUNT_MULS
+    unsigned long long z = 0;
+    return hipMemcpyToSymbol(HIP_SYMBOL(g_fp_mul_count), &z, 8) == hipSuccess ? 1 : 0;
+#else
+    return 0;
+#endif
+}
+extern "C" unsigned long long hbls_mulcount_read(void) {
+#ifdef HBLS_COUNT_MULS
+    unsigned long long v = 0;
+    if (hipMemcpyFromSymbol(&v, HIP_SYMBOL(g_fp_mul_count), 8) != hipSuccess) return 0;
+    return v;
+#else
+    return 0;
+#endif
 }
