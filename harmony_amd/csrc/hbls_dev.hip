@@ -1839,6 +1839,28 @@ extern "C" int hbls_sign_hash(const uint8_t sk32[32], const uint8_t *msg, size_t
 }
 
 /* ---- committee ---- */
+
+/* coop items-per-block dispatch: 16-item/64-thread blocks (75 KB arena,
+ * 2 blocks/CU) when the batch fills >= 2 blocks per CU; 8-item/32-thread
+ * blocks (38 KB arena, 4 blocks/CU) below, so mid/small batches get 2x the
+ * co-resident waves to hide dependent-mad stalls. */
+static int coop_items(size_t batch) {
+    static int force = -2;
+    if (force == -2) {
+        const char *e = getenv("HBLS_COOP_ITEMS");
+        force = e ? atoi(e) : 0;
+    }
+    if (force == 8 || force == 16) return force;
+    return batch >= 8192 ? 16 : 8;
+}
+
+#define LAUNCH_COOP(kern, batch, ...) do { \
+    int it_ = coop_items(batch); \
+    int nbc_ = (int)(((batch) + it_ - 1) / it_); \
+    if (it_ == 16) hipLaunchKernelGGL(kern<16>, dim3(nbc_), dim3(64), 0, 0, __VA_ARGS__); \
+    else hipLaunchKernelGGL(kern<8>, dim3(nbc_), dim3(32), 0, 0, __VA_ARGS__); \
+} while (0)
+
 struct hbls_committee {
     g1aff_t *d_table;
     g1_t *d_full_sum;   /* committee-wide key sum, for the dense-mask path */
@@ -1965,8 +1987,7 @@ static int run_agg_verify_pipeline(const hbls_committee_t *c, const uint8_t *d_b
     (void)hipEventRecord(ev[1], 0);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_hash_to_g2_coop, dim3(nbc), dim3(64), 0, 0,
-                           d_msg, (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+        LAUNCH_COOP(k_hash_to_g2_coop, batch, d_msg, (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
                            (int)batch, g_fast_cofactor);
     } else {
         hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
@@ -1976,8 +1997,7 @@ static int run_agg_verify_pipeline(const hbls_committee_t *c, const uint8_t *d_b
     (void)hipEventRecord(ev[2], 0);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_g2_decompress_coop, dim3(nbc), dim3(64), 0, 0,
-                           d_sig, dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+        LAUNCH_COOP(k_g2_decompress_coop, batch, d_sig, dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     } else {
         hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                            d_sig, dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
@@ -1985,8 +2005,7 @@ static int run_agg_verify_pipeline(const hbls_committee_t *c, const uint8_t *d_b
     (void)hipEventRecord(ev[3], 0);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_verify_coop, dim3(nbc), dim3(64), 0, 0,
-                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+        LAUNCH_COOP(k_verify_coop, batch, dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
     } else {
         hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0,
@@ -2048,8 +2067,7 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
     int nb = (int)((batch + 63) / 64);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_hash_to_g2_coop, dim3(nbc), dim3(64), 0, 0,
-                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+        LAUNCH_COOP(k_hash_to_g2_coop, batch, dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
                            (int)batch, g_fast_cofactor);
     } else {
         hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
@@ -2058,16 +2076,14 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
     }
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_g2_decompress_coop, dim3(nbc), dim3(64), 0, 0,
-                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+        LAUNCH_COOP(k_g2_decompress_coop, batch, dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     } else {
         hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                            dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     }
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_verify_votes_coop, dim3(nbc), dim3(64), 0, 0,
-                           c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
+        LAUNCH_COOP(k_verify_votes_coop, batch, c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
                            (const uint32_t *)nullptr,
                            dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
                            dres.as<int32_t>(), (int)batch);
@@ -2872,8 +2888,7 @@ extern "C" int hbls_batch_agg_verify_partials(
                            dpok.as<int32_t>(), (int)batch);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_hash_to_g2_coop, dim3(nbc), dim3(64), 0, 0,
-                           dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
+        LAUNCH_COOP(k_hash_to_g2_coop, batch, dmsg.as<uint8_t>(), (int)msg_len, dhm.as<g2_t>(), dhok.as<int32_t>(),
                            (int)batch, g_fast_cofactor);
     } else {
         hipLaunchKernelGGL(k_hash_to_g2, dim3(nb), dim3(64), 0, 0,
@@ -2882,16 +2897,14 @@ extern "C" int hbls_batch_agg_verify_partials(
     }
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_g2_decompress_coop, dim3(nbc), dim3(64), 0, 0,
-                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+        LAUNCH_COOP(k_g2_decompress_coop, batch, dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     } else {
         hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                            dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
     }
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_verify_coop, dim3(nbc), dim3(64), 0, 0,
-                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+        LAUNCH_COOP(k_verify_coop, batch, dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
     } else {
         hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0,
@@ -3203,8 +3216,7 @@ extern "C" int hbls_stream_set_rounds(hbls_stream *s, const uint32_t *slots, int
     HIP_OK(hipMemcpy(dpl.p, payloads, (size_t)k * plen, hipMemcpyHostToDevice));
     Timer tm;
     int nbc = (k + CV_ITEMS - 1) / CV_ITEMS;
-    hipLaunchKernelGGL(k_hash_to_g2_coop, dim3(nbc), dim3(64), 0, 0,
-                       dpl.as<uint8_t>(), (int)plen, dhm.as<g2_t>(), dok.as<int32_t>(),
+    LAUNCH_COOP(k_hash_to_g2_coop, (size_t)k, dpl.as<uint8_t>(), (int)plen, dhm.as<g2_t>(), dok.as<int32_t>(),
                        k, g_fast_cofactor);
     hipLaunchKernelGGL(k_stream_scatter_hm, dim3((k + 63) / 64), dim3(64), 0, 0,
                        s->d_hm, s->d_hm_ok, dhm.as<g2_t>(), dok.as<int32_t>(),
@@ -3248,10 +3260,8 @@ extern "C" int hbls_stream_process(hbls_stream *s, const uint32_t *key_idx,
                        dclamp.as<uint32_t>(), dpok.as<int32_t>(), (int)batch);
     if ((int)batch <= coop_threshold()) {
         int nbc = (int)((batch + CV_ITEMS - 1) / CV_ITEMS);
-        hipLaunchKernelGGL(k_g2_decompress_coop, dim3(nbc), dim3(64), 0, 0,
-                           dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
-        hipLaunchKernelGGL(k_verify_votes_coop, dim3(nbc), dim3(64), 0, 0,
-                           s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
+        LAUNCH_COOP(k_g2_decompress_coop, batch, dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
+        LAUNCH_COOP(k_verify_votes_coop, batch, s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
                            s->d_hm, dclamp.as<uint32_t>(),
                            dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), s->d_hm_ok,
                            dres.as<int32_t>(), (int)batch);
@@ -3305,8 +3315,7 @@ extern "C" int hbls_stream_check(hbls_stream *s, const uint32_t *slots, int k,
                           s->c->d_wtab, s->c->d_winf);
     if (k <= coop_threshold()) {
         int nbc = (k + CV_ITEMS - 1) / CV_ITEMS;
-        hipLaunchKernelGGL(k_verify_coop, dim3(nbc), dim3(64), 0, 0,
-                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+        LAUNCH_COOP(k_verify_coop, (size_t)k, dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), k);
     } else {
         hipLaunchKernelGGL(k_verify, dim3((k + 63) / 64), dim3(64), 0, 0,
