@@ -1151,6 +1151,8 @@ DEVN int verify_pairing(const g1_t &pub, const g2_t &hm, const g2aff_t &sig_aff,
     return fp12_is_one(f) ? 1 : 0;
 }
 
+#include "hbls_rf.inc"
+
 /* ================================================================ Keccak-256 */
 __constant__ uint64_t D_KECCAK_RC[24] = {
     0x0000000000000001ULL, 0x0000000000008082ULL, 0x800000000000808aULL, 0x8000000080008000ULL,
@@ -1840,6 +1842,37 @@ extern "C" int hbls_sign_hash(const uint8_t sk32[32], const uint8_t *msg, size_t
 
 /* ---- committee ---- */
 
+
+#ifndef HBLS_RF_DEFAULT
+#define HBLS_RF_DEFAULT 0
+#endif
+/* register-file verify dispatch (HBLS_VERIFY_RF: 0 round-1 kernel,
+ * 1 rf compiler-occupancy, 2 rf 256-reg, 3 rf 512-reg) */
+static int g_rf_override = -1;
+static int rf_mode(void) {
+    if (g_rf_override >= 0) return g_rf_override;
+    static int m = -1;
+    if (m < 0) { const char *e = getenv("HBLS_VERIFY_RF"); m = e ? atoi(e) : HBLS_RF_DEFAULT; }
+    return m;
+}
+extern "C" void hbls_set_verify_rf(int mode) { g_rf_override = mode; }
+#define LAUNCH_VERIFY_SCALAR(nb, ...) do { \
+    switch (rf_mode()) { \
+    case 1: hipLaunchKernelGGL(k_verify_rf, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    case 2: hipLaunchKernelGGL(k_verify_rf_w2, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    case 3: hipLaunchKernelGGL(k_verify_rf_w1, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    default: hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    } \
+} while (0)
+#define LAUNCH_VOTES_SCALAR(nb, ...) do { \
+    switch (rf_mode()) { \
+    case 1: hipLaunchKernelGGL(k_verify_votes_rf, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    case 2: hipLaunchKernelGGL(k_verify_votes_rf_w2, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    case 3: hipLaunchKernelGGL(k_verify_votes_rf_w1, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    default: hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    } \
+} while (0)
+
 /* coop items-per-block dispatch: 16-item/64-thread blocks (75 KB arena,
  * 2 blocks/CU) when the batch fills >= 2 blocks per CU; 8-item/32-thread
  * blocks (38 KB arena, 4 blocks/CU) below, so mid/small batches get 2x the
@@ -2008,8 +2041,7 @@ static int run_agg_verify_pipeline(const hbls_committee_t *c, const uint8_t *d_b
         LAUNCH_COOP(k_verify_coop, batch, dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
     } else {
-        hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0,
-                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+        LAUNCH_VERIFY_SCALAR(nb, dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
     }
     (void)hipEventRecord(ev[4], 0);
@@ -2088,8 +2120,7 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
                            dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
                            dres.as<int32_t>(), (int)batch);
     } else {
-        hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
-                           c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
+        LAUNCH_VOTES_SCALAR(nb, c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
                            (const uint32_t *)nullptr,
                            dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
                            dres.as<int32_t>(), (int)batch);
@@ -2907,8 +2938,7 @@ extern "C" int hbls_batch_agg_verify_partials(
         LAUNCH_COOP(k_verify_coop, batch, dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
     } else {
-        hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0,
-                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+        LAUNCH_VERIFY_SCALAR(nb, dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch);
     }
     if (n_ext)
@@ -3268,8 +3298,7 @@ extern "C" int hbls_stream_process(hbls_stream *s, const uint32_t *key_idx,
     } else {
         hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                            dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
-        hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
-                           s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
+        LAUNCH_VOTES_SCALAR(nb, s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
                            s->d_hm, dclamp.as<uint32_t>(),
                            dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), s->d_hm_ok,
                            dres.as<int32_t>(), (int)batch);
@@ -3318,8 +3347,7 @@ extern "C" int hbls_stream_check(hbls_stream *s, const uint32_t *slots, int k,
         LAUNCH_COOP(k_verify_coop, (size_t)k, dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), k);
     } else {
-        hipLaunchKernelGGL(k_verify, dim3((k + 63) / 64), dim3(64), 0, 0,
-                           dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
+        LAUNCH_VERIFY_SCALAR((k + 63) / 64, dagg.as<g1_t>(), dhm.as<g2_t>(), dsaff.as<g2aff_t>(),
                            dsflags.as<int32_t>(), dhok.as<int32_t>(), dres.as<int32_t>(), k);
     }
     tm.stop_and_store();

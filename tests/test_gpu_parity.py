@@ -528,3 +528,35 @@ def test_msm_pippenger_4096(core, capi):
     bad_sc[0:32] = b"\xff" * 32         # scalar >= r
     with _pytest.raises(ValueError):
         core.msm_g1(b"".join(pts), bytes(bad_sc), n)
+
+
+def test_verify_rf_parity(core, capi, keys16):
+    """Register-file verify kernels (hbls_rf.inc): accept/reject identical
+    to the round-1 kernel and the oracle across all three launch-bound
+    variants, incl. identity edges and wrong-signer rejects."""
+    sks, pks, n = keys16
+    gc = core.Committee(pks, n)
+    msg = pr.construct_commit_payload(77, pr.synth_msg(7), 9)
+    signers = [1, 2, 4, 8, 9, 13]
+    bm = bytearray((n + 7) // 8)
+    for i in signers:
+        bm[i >> 3] |= 1 << (i & 7)
+    agg = capi.aggregate_sigs([capi.sign_hash(sks[32 * i:32 * i + 32], msg)
+                               for i in signers])
+    bad = bytearray(bm)
+    bad[0] ^= 1
+    core._lib.hbls_set_verify_rf.argtypes = [__import__("ctypes").c_int]
+    try:
+        for mode in (0, 1, 2, 3):
+            core._lib.hbls_set_verify_rf(mode)
+            core.set_coop_threshold(0)        # force the scalar (rf) kernel
+            try:
+                assert gc.agg_verify(bytes(bm), agg, msg) is True, mode
+                assert gc.agg_verify(bytes(bad), agg, msg) is False, mode
+                # herumi identity edge through the rf path
+                assert gc.agg_verify(bytes((n + 7) // 8), b"\x00" * 96, msg) is True, mode
+                assert gc.agg_verify(bytes((n + 7) // 8), agg, msg) is False, mode
+            finally:
+                core.set_coop_threshold(-1)
+    finally:
+        core._lib.hbls_set_verify_rf(-1)
