@@ -389,20 +389,20 @@ def main():
     if rank != 0:
         return
 
-    # ---- algorithmic work accounting (oracle op counter, same config) ----
-    bm0 = bitmaps_cat[:bmlen]
-    capi.reset_op_count()
-    oc = capi.Committee(pks, COMMITTEE)
-    capi.reset_op_count()
-    oc.mask_aggregate(bm0)
-    f_mask = capi.op_count()
-    capi.reset_op_count()
-    capi.hash_to_g2(msgs[0])
-    f_hash = capi.op_count()
-    capi.reset_op_count()
-    oc.agg_verify(bm0, sigs[:96], msgs[0])
-    f_total = capi.op_count()
-    f_verify_stage = f_total - f_mask - f_hash   # decompress + pairing legs
+    # ---- algorithmic work accounting: the GPU path's OWN fp_mul counts,
+    # measured per operation by the instrumented build (libhbls_count.so,
+    # tools/count_muls.py on an MI355X -> gpurun_out/r2d_mulcounts.json).
+    # Round 1 used the oracle's op counter, which performs ~1.47x more muls
+    # than the GPU path (oracle 54,358 vs GPU 37,044 per aggregate-verify at
+    # this config) and overstated `achieved`; these are the honest GPU-side
+    # constants for committee=4096 Bernoulli(0.9) masks.
+    oc = capi.Committee(pks, COMMITTEE)      # oracle committee (cpu_baseline)
+    f_mask = 3829
+    f_hash = 7007
+    f_decompress = 4306
+    f_pairing = 21902
+    f_verify_stage = f_pairing               # the k_verify launch alone
+    f_total = 37044
 
     # ---- roofline: dominant stage = k_verify (pairing); live HIP-event time
     peak = core._lib.hbls_mad_peak_ops()
@@ -425,7 +425,9 @@ def main():
             "verify_pairing": round(stage_ns[3] / 1e6 / args.steps, 3),
         },
         "fp_muls_per_verify": {"mask": f_mask, "hash": f_hash,
-                               "verify_stage": f_verify_stage, "total": f_total},
+                               "decompress": f_decompress,
+                               "verify_stage": f_verify_stage, "total": f_total,
+                               "source": "GPU-instrumented (r2d_mulcounts)"},
     }
 
     # ---- CPU baseline: the oracle ("port"), OpenMP over items.  Two-phase:
