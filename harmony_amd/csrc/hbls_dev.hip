@@ -2231,6 +2231,69 @@ extern "C" int hbls_g2_check(const uint8_t p96[96]) {
     return ok ? HBLS_OK : HBLS_FALSE;
 }
 
+/* round-1 per-thread double-and-add MSM, kept ONLY as the A/B reference for
+ * the Pippenger kernel (hbls_msm_g1_naive; never on the product path) */
+__global__ void __launch_bounds__(256)
+k_msm_naive(const uint8_t *points48, const uint8_t *scalars32, int n,
+            g1_t *partials, int32_t *ok) {
+    __shared__ g1_t red[256];
+    int i = blockIdx.x * 256 + threadIdx.x;
+    g1_t acc;
+    g1_set_inf(acc);
+    if (i < n) {
+        g1_t p;
+        uint64_t k[4];
+        if (!g1_deserialize(p, points48 + (size_t)i * 48, true) ||
+            !fr_from_le32(k, scalars32 + (size_t)i * 32)) {
+            atomicExch(ok, 0);
+        } else {
+            g1_mul(acc, p, k, 4);
+        }
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int s = 128; s > 0; s >>= 1) {
+        if (threadIdx.x < s) {
+            g1_t t;
+            g1_add(t, red[threadIdx.x], red[threadIdx.x + s]);
+            red[threadIdx.x] = t;
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) partials[blockIdx.x] = red[0];
+}
+__global__ void k_g1_reduce_seq(const g1_t *partials, int n, uint8_t *out48) {
+    g1_t acc;
+    g1_set_inf(acc);
+    for (int i = 0; i < n; i++) g1_add(acc, acc, partials[i]);
+    g1_serialize(out48, acc);
+}
+extern "C" int hbls_msm_g1_naive(const uint8_t *points48, const uint8_t *scalars32,
+                                 size_t n, uint8_t out48[48]) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    int nblocks = (int)((n + 255) / 256);
+    DevBuf dp(n * 48), ds(n * 32), dpart(nblocks * sizeof(g1_t)), dok(4), dout(48);
+    if (dp.err || ds.err || dpart.err || dok.err || dout.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(dp.p, points48, n * 48, hipMemcpyHostToDevice));
+    HIP_OK(hipMemcpy(ds.p, scalars32, n * 32, hipMemcpyHostToDevice));
+    int32_t one = 1;
+    HIP_OK(hipMemcpy(dok.p, &one, 4, hipMemcpyHostToDevice));
+    Timer tm;
+    hipLaunchKernelGGL(k_msm_naive, dim3(nblocks), dim3(256), 0, 0,
+                       dp.as<uint8_t>(), ds.as<uint8_t>(), (int)n,
+                       dpart.as<g1_t>(), dok.as<int32_t>());
+    hipLaunchKernelGGL(k_g1_reduce_seq, dim3(1), dim3(1), 0, 0,
+                       dpart.as<g1_t>(), nblocks, dout.as<uint8_t>());
+    tm.stop_and_store();
+    HIP_OK(hipGetLastError());
+    int32_t ok;
+    HIP_OK(hipMemcpy(&ok, dok.p, 4, hipMemcpyDeviceToHost));
+    if (!ok) return HBLS_ERR_BADINPUT;
+    HIP_OK(hipMemcpy(out48, dout.p, 48, hipMemcpyDeviceToHost));
+    return HBLS_OK;
+}
+
 /* ---- Pippenger bucket MSM (the `north_star`'s general-scalar MSM) ----
  * radix-256 digits (c = 8): 32 windows x 255 buckets.  GPU shape:
  *   prep     — thread/point: decompress + subgroup-check + digit transpose
