@@ -1846,6 +1846,13 @@ extern "C" int hbls_sign_hash(const uint8_t sk32[32], const uint8_t *msg, size_t
 #ifndef HBLS_RF_DEFAULT
 #define HBLS_RF_DEFAULT 0
 #endif
+#ifdef HBLS_RF_ALL_LB
+#define HBLS_RF_W2_KERN k_verify_rf_w2
+#define HBLS_RF_W1_KERN k_verify_rf_w1
+#else
+#define HBLS_RF_W2_KERN k_verify_rf
+#define HBLS_RF_W1_KERN k_verify_rf
+#endif
 /* register-file verify dispatch (HBLS_VERIFY_RF: 0 round-1 kernel,
  * 1 rf compiler-occupancy, 2 rf 256-reg, 3 rf 512-reg) */
 static int g_rf_override = -1;
@@ -1859,18 +1866,20 @@ extern "C" void hbls_set_verify_rf(int mode) { g_rf_override = mode; }
 #define LAUNCH_VERIFY_SCALAR(nb, ...) do { \
     switch (rf_mode()) { \
     case 1: hipLaunchKernelGGL(k_verify_rf, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
-    case 2: hipLaunchKernelGGL(k_verify_rf_w2, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
-    case 3: hipLaunchKernelGGL(k_verify_rf_w1, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    case 2: hipLaunchKernelGGL(HBLS_RF_W2_KERN, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    case 3: hipLaunchKernelGGL(HBLS_RF_W1_KERN, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
     default: hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
     } \
 } while (0)
-#define LAUNCH_VOTES_SCALAR(nb, ...) do { \
-    switch (rf_mode()) { \
-    case 1: hipLaunchKernelGGL(k_verify_votes_rf, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
-    case 2: hipLaunchKernelGGL(k_verify_votes_rf_w2, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
-    case 3: hipLaunchKernelGGL(k_verify_votes_rf_w1, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
-    default: hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
-    } \
+/* votes through the rf path go: gather (pubs + per-item hm) -> k_verify_rf.
+ * gb_pub/gb_hm/gb_ok are caller-allocated DevBufs sized batch. */
+#define LAUNCH_VOTES_SCALAR_RF(nb, table, nn, didx, dhm, dhmidx, dsaff, dsflags, dhok, dres, batch, gb_pub, gb_hm, gb_ok, RFKERN) do { \
+    hipLaunchKernelGGL(k_gather_votes_rf, dim3(nb), dim3(64), 0, 0, \
+                       table, nn, didx, dhm, dhmidx, dhok, \
+                       (gb_pub).as<g1_t>(), (gb_hm).as<g2_t>(), (gb_ok).as<int32_t>(), batch); \
+    hipLaunchKernelGGL(RFKERN, dim3(nb), dim3(64), 0, 0, \
+                       (gb_pub).as<g1_t>(), (gb_hm).as<g2_t>(), dsaff, dsflags, \
+                       (gb_ok).as<int32_t>(), dres, batch); \
 } while (0)
 
 /* coop items-per-block dispatch: 16-item/64-thread blocks (75 KB arena,
@@ -2090,8 +2099,10 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
     DevBuf didx(batch * 4), dsig(batch * 96), dmsg(batch * msg_len);
     DevBuf dhm(batch * sizeof(g2_t)), dsaff(batch * sizeof(g2aff_t));
     DevBuf dsflags(batch * 4), dhok(batch * 4), dres(batch * 4);
+    size_t gsz = rf_mode() ? batch : 1;   /* rf votes gather buffers */
+    DevBuf gpub(gsz * sizeof(g1_t)), ghm(gsz * sizeof(g2_t)), gok(gsz * 4);
     if (didx.err || dsig.err || dmsg.err || dhm.err || dsaff.err || dsflags.err ||
-        dhok.err || dres.err) return HBLS_ERR;
+        dhok.err || dres.err || gpub.err || ghm.err || gok.err) return HBLS_ERR;
     HIP_OK(hipMemcpy(didx.p, key_idx, batch * 4, hipMemcpyHostToDevice));
     HIP_OK(hipMemcpy(dsig.p, sigs96, batch * 96, hipMemcpyHostToDevice));
     HIP_OK(hipMemcpy(dmsg.p, msgs, batch * msg_len, hipMemcpyHostToDevice));
@@ -2120,10 +2131,19 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
                            dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
                            dres.as<int32_t>(), (int)batch);
     } else {
-        LAUNCH_VOTES_SCALAR(nb, c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
-                           (const uint32_t *)nullptr,
-                           dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
-                           dres.as<int32_t>(), (int)batch);
+        if (rf_mode()) {
+            LAUNCH_VOTES_SCALAR_RF(nb, c->d_table, (int)c->n, didx.as<uint32_t>(),
+                                   dhm.as<g2_t>(), (const uint32_t *)nullptr,
+                                   dsaff.as<g2aff_t>(), dsflags.as<int32_t>(),
+                                   dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch,
+                                   gpub, ghm, gok, k_verify_rf);
+        } else {
+            hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
+                               c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
+                               (const uint32_t *)nullptr,
+                               dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
+                               dres.as<int32_t>(), (int)batch);
+        }
     }
     tm.stop_and_store();
     HIP_OK(hipGetLastError());
@@ -3340,8 +3360,11 @@ extern "C" int hbls_stream_process(hbls_stream *s, const uint32_t *key_idx,
     DevBuf dclamp(batch * 4), dpok(batch * 4);
     DevBuf dsaff(batch * sizeof(g2aff_t)), dsflags(batch * 4), dres(batch * 4);
     DevBuf dact(n_active * 4);
+    size_t gsz = rf_mode() ? batch : 1;   /* rf votes gather buffers */
+    DevBuf gpub(gsz * sizeof(g1_t)), ghm(gsz * sizeof(g2_t)), gok(gsz * 4);
     if (didx.err || dridx.err || dsig.err || dclamp.err || dpok.err ||
-        dsaff.err || dsflags.err || dres.err || dact.err) return HBLS_ERR;
+        dsaff.err || dsflags.err || dres.err || dact.err ||
+        gpub.err || ghm.err || gok.err) return HBLS_ERR;
     HIP_OK(hipMemcpy(didx.p, key_idx, batch * 4, hipMemcpyHostToDevice));
     HIP_OK(hipMemcpy(dridx.p, round_idx, batch * 4, hipMemcpyHostToDevice));
     HIP_OK(hipMemcpy(dsig.p, sigs96, batch * 96, hipMemcpyHostToDevice));
@@ -3361,10 +3384,19 @@ extern "C" int hbls_stream_process(hbls_stream *s, const uint32_t *key_idx,
     } else {
         hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                            dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
-        LAUNCH_VOTES_SCALAR(nb, s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
-                           s->d_hm, dclamp.as<uint32_t>(),
-                           dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), s->d_hm_ok,
-                           dres.as<int32_t>(), (int)batch);
+        if (rf_mode()) {
+            LAUNCH_VOTES_SCALAR_RF(nb, s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
+                                   s->d_hm, dclamp.as<uint32_t>(),
+                                   dsaff.as<g2aff_t>(), dsflags.as<int32_t>(),
+                                   s->d_hm_ok, dres.as<int32_t>(), (int)batch,
+                                   gpub, ghm, gok, k_verify_rf);
+        } else {
+            hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
+                               s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
+                               s->d_hm, dclamp.as<uint32_t>(),
+                               dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), s->d_hm_ok,
+                               dres.as<int32_t>(), (int)batch);
+        }
     }
     hipLaunchKernelGGL(k_merge_pok, dim3(nb), dim3(64), 0, 0,
                        dres.as<int32_t>(), dpok.as<int32_t>(), (int)batch);
