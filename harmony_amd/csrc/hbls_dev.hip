@@ -1882,10 +1882,12 @@ extern "C" void hbls_set_verify_rf(int mode) { g_rf_override = mode; }
                        (gb_ok).as<int32_t>(), dres, batch); \
 } while (0)
 
-/* coop items-per-block dispatch: 16-item/64-thread blocks (75 KB arena,
- * 2 blocks/CU) when the batch fills >= 2 blocks per CU; 8-item/32-thread
- * blocks (38 KB arena, 4 blocks/CU) below, so mid/small batches get 2x the
- * co-resident waves to hide dependent-mad stalls. */
+/* coop items-per-block dispatch.  MEASURED (gpurun_out/r2g_coopab.log):
+ * the 8-item/32-thread variant (38 KB arena, 4 blocks/CU) is WORSE at every
+ * batch (e.g. 116 vs 82 ms at 4096, stream 17.9k vs 19.2k msgs/s) — the
+ * half-empty waves double the per-item instruction issue, which costs more
+ * than the extra co-residency hides.  Default is therefore 16 items
+ * everywhere; HBLS_COOP_ITEMS=8 keeps the losing variant reproducible. */
 static int coop_items(size_t batch) {
     static int force = -2;
     if (force == -2) {
@@ -1893,7 +1895,8 @@ static int coop_items(size_t batch) {
         force = e ? atoi(e) : 0;
     }
     if (force == 8 || force == 16) return force;
-    return batch >= 8192 ? 16 : 8;
+    (void)batch;
+    return 16;
 }
 
 #define LAUNCH_COOP(kern, batch, ...) do { \
