@@ -1151,7 +1151,48 @@ DEVN int verify_pairing(const g1_t &pub, const g2_t &hm, const g2aff_t &sig_aff,
     return fp12_is_one(f) ? 1 : 0;
 }
 
+/* split-leg pairing: two single-leg Miller loops (f = f1 * f2) instead of
+ * the fused shared-squaring loop.  +63 fp12_sqr of work (~10%) but the live
+ * state drops by T2/Q2/P2 (~680 B/thread of hot scratch) — an A/B against
+ * the scratch-bandwidth bound (HBLS_VERIFY_RF=4). */
+DEVN int verify_pairing_2p(const g1_t &pub, const g2_t &hm, const g2aff_t &sig_aff,
+                           bool sig_inf) {
+    bool pub_inf = g1_is_inf(pub);
+    if (pub_inf && sig_inf) return 1;
+    if (pub_inf || sig_inf) return 0;
+    g1aff_t pa, ba;
+    g2aff_t ha;
+    g1_to_affine(pa, pub);
+    { g1_t base, nb;
+      fp_load(base.x, BLS_G1_X); fp_load(base.y, BLS_G1_Y); fp_one(base.z);
+      g1_neg(nb, base);
+      ba.x = nb.x; ba.y = nb.y; }
+    g2_t hmc = hm;
+    g2_to_affine(ha, hmc);
+    fp12_t f1, f2;
+    fp12_one(f1);
+    miller_loop_acc(f1, ha, pa);
+    fp12_one(f2);
+    miller_loop_acc(f2, sig_aff, ba);
+    fp12_mul(f1, f1, f2);
+    fp12_conj(f1, f1);
+    final_exp(f1, f1);
+    return fp12_is_one(f1) ? 1 : 0;
+}
+__global__ void k_verify_2p(const g1_t *aggpubs, const g2_t *hms, const g2aff_t *sigs,
+                            const int32_t *sig_flags, const int32_t *hm_ok,
+                            int32_t *results, int batch) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= batch) return;
+    if (!hm_ok[i] || sig_flags[i] == 0) { results[i] = HBLS_ERR_BADINPUT; return; }
+    g2aff_t dummy;
+    bool sig_inf = sig_flags[i] == 2;
+    results[i] = verify_pairing_2p(aggpubs[i], hms[i], sig_inf ? dummy : sigs[i], sig_inf);
+}
+
+#ifdef HBLS_RF
 #include "hbls_rf.inc"
+#endif
 
 /* ================================================================ Keccak-256 */
 __constant__ uint64_t D_KECCAK_RC[24] = {
@@ -1863,11 +1904,18 @@ static int rf_mode(void) {
     return m;
 }
 extern "C" void hbls_set_verify_rf(int mode) { g_rf_override = mode; }
-#define LAUNCH_VERIFY_SCALAR(nb, ...) do { \
-    switch (rf_mode()) { \
+#ifdef HBLS_RF
+#define HBLS_RF_CASES(nb, ...) \
     case 1: hipLaunchKernelGGL(k_verify_rf, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
     case 2: hipLaunchKernelGGL(HBLS_RF_W2_KERN, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
-    case 3: hipLaunchKernelGGL(HBLS_RF_W1_KERN, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    case 3: hipLaunchKernelGGL(HBLS_RF_W1_KERN, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break;
+#else
+#define HBLS_RF_CASES(nb, ...)
+#endif
+#define LAUNCH_VERIFY_SCALAR(nb, ...) do { \
+    switch (rf_mode()) { \
+    HBLS_RF_CASES(nb, __VA_ARGS__) \
+    case 4: hipLaunchKernelGGL(k_verify_2p, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
     default: hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
     } \
 } while (0)
@@ -2134,13 +2182,16 @@ extern "C" int hbls_batch_verify_votes(const hbls_committee_t *c, const uint32_t
                            dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), dhok.as<int32_t>(),
                            dres.as<int32_t>(), (int)batch);
     } else {
-        if (rf_mode()) {
+#ifdef HBLS_RF
+        if (rf_mode() >= 1 && rf_mode() <= 3) {
             LAUNCH_VOTES_SCALAR_RF(nb, c->d_table, (int)c->n, didx.as<uint32_t>(),
                                    dhm.as<g2_t>(), (const uint32_t *)nullptr,
                                    dsaff.as<g2aff_t>(), dsflags.as<int32_t>(),
                                    dhok.as<int32_t>(), dres.as<int32_t>(), (int)batch,
                                    gpub, ghm, gok, k_verify_rf);
-        } else {
+        } else
+#endif
+        {
             hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
                                c->d_table, (int)c->n, didx.as<uint32_t>(), dhm.as<g2_t>(),
                                (const uint32_t *)nullptr,
@@ -3387,13 +3438,16 @@ extern "C" int hbls_stream_process(hbls_stream *s, const uint32_t *key_idx,
     } else {
         hipLaunchKernelGGL(k_g2_decompress, dim3(nb), dim3(64), 0, 0,
                            dsig.as<uint8_t>(), dsaff.as<g2aff_t>(), dsflags.as<int32_t>(), (int)batch);
-        if (rf_mode()) {
+#ifdef HBLS_RF
+        if (rf_mode() >= 1 && rf_mode() <= 3) {
             LAUNCH_VOTES_SCALAR_RF(nb, s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
                                    s->d_hm, dclamp.as<uint32_t>(),
                                    dsaff.as<g2aff_t>(), dsflags.as<int32_t>(),
                                    s->d_hm_ok, dres.as<int32_t>(), (int)batch,
                                    gpub, ghm, gok, k_verify_rf);
-        } else {
+        } else
+#endif
+        {
             hipLaunchKernelGGL(k_verify_votes, dim3(nb), dim3(64), 0, 0,
                                s->c->d_table, (int)s->c->n, didx.as<uint32_t>(),
                                s->d_hm, dclamp.as<uint32_t>(),
@@ -3488,6 +3542,14 @@ extern "C" unsigned long long hbls_mulcount_read(void) {
     unsigned long long v = 0;
     if (hipMemcpyFromSymbol(&v, HIP_SYMBOL(g_fp_mul_count), 8) != hipSuccess) return 0;
     return v;
+#else
+    return 0;
+#endif
+}
+
+extern "C" int hbls_has_rf(void) {
+#ifdef HBLS_RF
+    return 1;
 #else
     return 0;
 #endif

@@ -531,9 +531,10 @@ def test_msm_pippenger_4096(core, capi):
 
 
 def test_verify_rf_parity(core, capi, keys16):
-    """Register-file verify kernels (hbls_rf.inc): accept/reject identical
-    to the round-1 kernel and the oracle across all three launch-bound
-    variants, incl. identity edges and wrong-signer rejects."""
+    """Alternative verify kernels: the split-leg pairing (mode 4, always
+    built) and — when the opt-in -DHBLS_RF build is present — the
+    register-file kernels, accept/reject identical to the round-1 kernel
+    incl. identity edges and wrong-signer rejects."""
     sks, pks, n = keys16
     gc = core.Committee(pks, n)
     msg = pr.construct_commit_payload(77, pr.synth_msg(7), 9)
@@ -546,8 +547,9 @@ def test_verify_rf_parity(core, capi, keys16):
     bad = bytearray(bm)
     bad[0] ^= 1
     core._lib.hbls_set_verify_rf.argtypes = [__import__("ctypes").c_int]
+    modes = (0, 4) if not core._lib.hbls_has_rf() else (0, 1, 2, 3, 4)
     try:
-        for mode in (0, 1, 2, 3):
+        for mode in modes:
             core._lib.hbls_set_verify_rf(mode)
             core.set_coop_threshold(0)        # force the scalar (rf) kernel
             try:
