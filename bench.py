@@ -415,9 +415,9 @@ def main():
         "unit": "Gmad64/s",   # 64x64->128 multiply-accumulates (measured peak)
         "frac": round(ach / peak, 4) if peak else None,
         # memory-side bytes per k_verify launch, from the committed PMC passes
-        # (profiles/r01c_pmc.txt: FETCH+WRITE ~= 3.01 MB per verify, scratch-
+        # (profiles/r02b_pmc.txt: FETCH+WRITE = 2.81 MB per verify, scratch-
         # dominated; FETCH is a lower bound on gfx950 — see the profile header)
-        "traffic": int(3.01e6 * args.batch),
+        "traffic": int(2.81e6 * args.batch),
         "stages_ms_per_launch": {
             "mask_aggregate": round(stage_ns[0] / 1e6 / args.steps, 3),
             "hash_to_g2": round(stage_ns[1] / 1e6 / args.steps, 3),
