@@ -320,7 +320,7 @@ def main():
 
     # CPU-baseline thread count: the GPU boxes cgroup-cap this container at
     # 16 CPUs (cpu.max 1600000/100000) over 256 SMT threads; the measured
-    # sweep (gpurun_out/r2a_cpusweep.json) peaks at 64 OMP
+    # sweep (profiles/r02_data/r2a_cpusweep.json) peaks at 64 OMP
     # threads and COLLAPSES >=128 (oversubscription thrash under the quota).
     os.environ.setdefault("OMP_NUM_THREADS", str(min(64, os.cpu_count() or 64)))
     from oracle import capi, pyref as pr
@@ -391,7 +391,7 @@ def main():
 
     # ---- algorithmic work accounting: the GPU path's OWN fp_mul counts,
     # measured per operation by the instrumented build (libhbls_count.so,
-    # tools/count_muls.py on an MI355X -> gpurun_out/r2d_mulcounts.json).
+    # tools/count_muls.py on an MI355X -> profiles/r02_data/r2d_mulcounts.json).
     # Round 1 used the oracle's op counter, which performs ~1.47x more muls
     # than the GPU path (oracle 54,358 vs GPU 37,044 per aggregate-verify at
     # this config) and overstated `achieved`; these are the honest GPU-side

@@ -1930,7 +1930,7 @@ extern "C" void hbls_set_verify_rf(int mode) { g_rf_override = mode; }
                        (gb_ok).as<int32_t>(), dres, batch); \
 } while (0)
 
-/* coop items-per-block dispatch.  MEASURED (gpurun_out/r2g_coopab.log):
+/* coop items-per-block dispatch.  MEASURED (profiles/r02_data/r2g_coopab.log):
  * the 8-item/32-thread variant (38 KB arena, 4 blocks/CU) is WORSE at every
  * batch (e.g. 116 vs 82 ms at 4096, stream 17.9k vs 19.2k msgs/s) — the
  * half-empty waves double the per-item instruction issue, which costs more
