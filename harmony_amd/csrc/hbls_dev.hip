@@ -2402,15 +2402,49 @@ __global__ void k_msm_prep(const uint8_t *points48, const uint8_t *scalars32, in
     }
 }
 
+/* counting sort of (window, digit) -> point-index lists, so each bucket
+ * wave touches exactly its own points: no digit scans, no divergence
+ * (the first bucket design scanned all n digits per wave and lost to the
+ * naive kernel — profiles/r02_data/r2i_msm_ab.json) */
+__global__ void k_msm_hist(const uint8_t *digits, int n, uint32_t *cnt /* 32*256 */) {
+    size_t total = (size_t)32 * n;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t t = (size_t)blockIdx.x * blockDim.x + threadIdx.x; t < total; t += stride)
+        atomicAdd(&cnt[(t / n) * 256 + digits[t]], 1u);
+}
+__global__ void __launch_bounds__(64) k_msm_prefix(const uint32_t *cnt,
+                                                   uint32_t *off /* 32*257 */,
+                                                   uint32_t *cursor /* 32*256 */) {
+    int w = blockIdx.x * blockDim.x + threadIdx.x;
+    if (w >= 32) return;
+    uint32_t acc = 0;
+    for (int b = 0; b < 256; b++) {
+        off[w * 257 + b] = acc;
+        cursor[w * 256 + b] = acc;
+        acc += cnt[w * 256 + b];
+    }
+    off[w * 257 + 256] = acc;
+}
+__global__ void k_msm_scatter(const uint8_t *digits, int n, uint32_t *cursor,
+                              uint32_t *list /* 32*n */) {
+    size_t total = (size_t)32 * n;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t t = (size_t)blockIdx.x * blockDim.x + threadIdx.x; t < total; t += stride) {
+        size_t w = t / n;
+        uint32_t slot = atomicAdd(&cursor[w * 256 + digits[t]], 1u);
+        list[w * n + slot] = (uint32_t)(t - w * n);
+    }
+}
 __global__ void __launch_bounds__(64) k_msm_buckets(
-        const g1aff_t *pts, const uint8_t *digits, int n, g1_t *buckets /* 32*255 */) {
+        const g1aff_t *pts, const uint32_t *list, const uint32_t *off, int n,
+        g1_t *buckets /* 32*255 */) {
     int w = blockIdx.x / 255;
     int b = blockIdx.x % 255 + 1;
-    const uint8_t *row = digits + (size_t)w * n;
+    uint32_t lo = off[w * 257 + b], hi = off[w * 257 + b + 1];
     g1_t acc;
     g1_set_inf(acc);
-    for (int i = threadIdx.x; i < n; i += 64)
-        if (row[i] == b) g1_madd_i(acc, acc, pts[i]);
+    for (uint32_t k = lo + threadIdx.x; k < hi; k += 64)
+        g1_madd_i(acc, acc, pts[list[(size_t)w * n + k]]);
     __shared__ g1_t red[64];
     red[threadIdx.x] = acc;
     __syncthreads();
@@ -2462,19 +2496,31 @@ extern "C" int hbls_msm_g1(const uint8_t *points48, const uint8_t *scalars32, si
     if (rc != HBLS_OK) return rc;
     if (n == 0) return HBLS_ERR_BADINPUT;
     DevBuf dp(n * 48), ds(n * 32), dpts(n * sizeof(g1aff_t)), ddig(n * 32);
+    DevBuf dcnt(32 * 256 * 4), doff(32 * 257 * 4), dcur(32 * 256 * 4), dlist(n * 32 * 4);
     DevBuf dbuck(32 * 255 * sizeof(g1_t)), dws(32 * sizeof(g1_t)), dok(4), dout(48);
-    if (dp.err || ds.err || dpts.err || ddig.err || dbuck.err || dws.err ||
+    if (dp.err || ds.err || dpts.err || ddig.err || dcnt.err || doff.err ||
+        dcur.err || dlist.err || dbuck.err || dws.err ||
         dok.err || dout.err) return HBLS_ERR;
     HIP_OK(hipMemcpy(dp.p, points48, n * 48, hipMemcpyHostToDevice));
     HIP_OK(hipMemcpy(ds.p, scalars32, n * 32, hipMemcpyHostToDevice));
     int32_t one = 1;
     HIP_OK(hipMemcpy(dok.p, &one, 4, hipMemcpyHostToDevice));
+    HIP_OK(hipMemset(dcnt.p, 0, 32 * 256 * 4));
     Timer tm;
     hipLaunchKernelGGL(k_msm_prep, dim3((uint32_t)((n + 63) / 64)), dim3(64), 0, 0,
                        dp.as<uint8_t>(), ds.as<uint8_t>(), (int)n,
                        dpts.as<g1aff_t>(), ddig.as<uint8_t>(), dok.as<int32_t>());
+    int nb_sc = (int)(((size_t)32 * n + 255) / 256);
+    if (nb_sc > 4096) nb_sc = 4096;
+    hipLaunchKernelGGL(k_msm_hist, dim3(nb_sc), dim3(256), 0, 0,
+                       ddig.as<uint8_t>(), (int)n, dcnt.as<uint32_t>());
+    hipLaunchKernelGGL(k_msm_prefix, dim3(1), dim3(64), 0, 0,
+                       dcnt.as<uint32_t>(), doff.as<uint32_t>(), dcur.as<uint32_t>());
+    hipLaunchKernelGGL(k_msm_scatter, dim3(nb_sc), dim3(256), 0, 0,
+                       ddig.as<uint8_t>(), (int)n, dcur.as<uint32_t>(), dlist.as<uint32_t>());
     hipLaunchKernelGGL(k_msm_buckets, dim3(32 * 255), dim3(64), 0, 0,
-                       dpts.as<g1aff_t>(), ddig.as<uint8_t>(), (int)n, dbuck.as<g1_t>());
+                       dpts.as<g1aff_t>(), dlist.as<uint32_t>(), doff.as<uint32_t>(),
+                       (int)n, dbuck.as<g1_t>());
     hipLaunchKernelGGL(k_msm_wreduce, dim3(1), dim3(64), 0, 0,
                        dbuck.as<g1_t>(), dws.as<g1_t>());
     hipLaunchKernelGGL(k_msm_combine, dim3(1), dim3(1), 0, 0,
