@@ -58,6 +58,8 @@ def _load():
         ctypes.POINTER(ctypes.c_int32)]
     lib.hbls_mask_partials.argtypes = [ctypes.c_void_p, ctypes.c_char_p,
                                        ctypes.c_size_t, ctypes.c_char_p]
+    lib.hbls_msm_g1_committee.argtypes = [ctypes.c_void_p, ctypes.c_char_p,
+                                          ctypes.c_char_p]
     lib.hbls_batch_agg_verify_partials.argtypes = [
         ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t,
         ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_size_t,
@@ -287,6 +289,14 @@ class Committee:
                                            msgs, mlen, batch, res),
                "batch_seal_verify")
         return list(res)
+
+    def msm(self, scalars_cat: bytes) -> bytes:
+        """Pippenger MSM over the resident validated table: sum s_i * pk_i"""
+        if len(scalars_cat) != 32 * self.n:
+            raise ValueError("msm: scalars must be n*32 bytes")
+        out = ctypes.create_string_buffer(48)
+        _check(_lib.hbls_msm_g1_committee(self._h, scalars_cat, out), "msm_committee")
+        return out.raw
 
     def batch_verify_votes(self, key_idx, sigs: bytes, msgs: bytes, mlen: int):
         batch = len(key_idx)

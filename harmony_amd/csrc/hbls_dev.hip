@@ -2490,6 +2490,79 @@ __global__ void k_msm_combine(const g1_t *wsums, uint8_t *out48) {
     g1_serialize(out48, acc);
 }
 
+/* digits-only prep against an already-validated resident point table
+ * (committee handle): the serialized-input entry below spends most of its
+ * time decompressing + subgroup-checking points (~4.5k fp-muls each, same
+ * cost the naive kernel pays) — production callers hold the committee
+ * table resident (UpdateParticipants, quorum.go:326-334), so the MSM core
+ * should be callable without re-validating points. */
+__global__ void k_msm_digits(const uint8_t *scalars32, int n, uint8_t *digits,
+                             int32_t *ok) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t k[4];
+    if (!fr_from_le32(k, scalars32 + (size_t)i * 32)) {
+        atomicExch(ok, 0);
+        return;
+    }
+    for (int w = 0; w < 32; w++)
+        digits[(size_t)w * n + i] = (uint8_t)(k[w >> 3] >> (8 * (w & 7)));
+}
+
+static int msm_core(const g1aff_t *d_pts, const uint8_t *d_digits, size_t n,
+                    uint8_t *d_out48) {
+    DevBuf dcnt(32 * 256 * 4), doff(32 * 257 * 4), dcur(32 * 256 * 4), dlist(n * 32 * 4);
+    DevBuf dbuck(32 * 255 * sizeof(g1_t)), dws(32 * sizeof(g1_t));
+    if (dcnt.err || doff.err || dcur.err || dlist.err || dbuck.err || dws.err)
+        return HBLS_ERR;
+    HIP_OK(hipMemset(dcnt.p, 0, 32 * 256 * 4));
+    int nb_sc = (int)(((size_t)32 * n + 255) / 256);
+    if (nb_sc > 4096) nb_sc = 4096;
+    hipLaunchKernelGGL(k_msm_hist, dim3(nb_sc), dim3(256), 0, 0,
+                       d_digits, (int)n, dcnt.as<uint32_t>());
+    hipLaunchKernelGGL(k_msm_prefix, dim3(1), dim3(64), 0, 0,
+                       dcnt.as<uint32_t>(), doff.as<uint32_t>(), dcur.as<uint32_t>());
+    hipLaunchKernelGGL(k_msm_scatter, dim3(nb_sc), dim3(256), 0, 0,
+                       d_digits, (int)n, dcur.as<uint32_t>(), dlist.as<uint32_t>());
+    hipLaunchKernelGGL(k_msm_buckets, dim3(32 * 255), dim3(64), 0, 0,
+                       d_pts, dlist.as<uint32_t>(), doff.as<uint32_t>(),
+                       (int)n, dbuck.as<g1_t>());
+    hipLaunchKernelGGL(k_msm_wreduce, dim3(1), dim3(64), 0, 0,
+                       dbuck.as<g1_t>(), dws.as<g1_t>());
+    hipLaunchKernelGGL(k_msm_combine, dim3(1), dim3(1), 0, 0,
+                       dws.as<g1_t>(), d_out48);
+    HIP_OK(hipGetLastError());
+    /* keep the intermediate buffers alive until the chain completes (hipFree
+     * in ~DevBuf synchronizes, but be explicit about the dependency) */
+    HIP_OK(hipDeviceSynchronize());
+    return HBLS_OK;
+}
+
+/* MSM against the resident committee table: out = sum scalar_i * table[i].
+ * Points were decompressed + subgroup-checked once at committee build. */
+extern "C" int hbls_msm_g1_committee(const hbls_committee_t *c, const uint8_t *scalars32,
+                                     uint8_t out48[48]) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    size_t n = c->n;
+    DevBuf ds(n * 32), ddig(n * 32), dok(4), dout(48);
+    if (ds.err || ddig.err || dok.err || dout.err) return HBLS_ERR;
+    HIP_OK(hipMemcpy(ds.p, scalars32, n * 32, hipMemcpyHostToDevice));
+    int32_t one = 1;
+    HIP_OK(hipMemcpy(dok.p, &one, 4, hipMemcpyHostToDevice));
+    Timer tm;
+    hipLaunchKernelGGL(k_msm_digits, dim3((uint32_t)((n + 63) / 64)), dim3(64), 0, 0,
+                       ds.as<uint8_t>(), (int)n, ddig.as<uint8_t>(), dok.as<int32_t>());
+    rc = msm_core(c->d_table, ddig.as<uint8_t>(), n, dout.as<uint8_t>());
+    tm.stop_and_store();
+    if (rc != HBLS_OK) return rc;
+    int32_t ok;
+    HIP_OK(hipMemcpy(&ok, dok.p, 4, hipMemcpyDeviceToHost));
+    if (!ok) return HBLS_ERR_BADINPUT;
+    HIP_OK(hipMemcpy(out48, dout.p, 48, hipMemcpyDeviceToHost));
+    return HBLS_OK;
+}
+
 extern "C" int hbls_msm_g1(const uint8_t *points48, const uint8_t *scalars32, size_t n,
                            uint8_t out48[48]) {
     int rc = require_gpu();

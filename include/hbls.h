@@ -142,10 +142,15 @@ int hbls_batch_agg_verify_partials(const hbls_committee_t *c, const uint8_t *bit
                                    const uint8_t *sigs96, const uint8_t *msgs,
                                    size_t msg_len, size_t batch, int32_t *results);
 
-/* G1 MSM: out = sum_i scalar_i * P_i (scalars 32B LE each; general building
- * block behind mask aggregation; Pippenger on device) */
+/* G1 MSM: out = sum_i scalar_i * P_i (scalars 32B LE each; Pippenger
+ * bucket accumulation on device).  The serialized-points entry validates
+ * every point (decompress + subgroup check); the committee entry runs the
+ * MSM core against the resident, already-validated table — the production
+ * shape (UpdateParticipants holds the table per epoch, quorum.go:326-334) */
 int hbls_msm_g1(const uint8_t *points48, const uint8_t *scalars32, size_t n,
                 uint8_t out48[48]);
+int hbls_msm_g1_committee(const hbls_committee_t *c, const uint8_t *scalars32,
+                          uint8_t out48[48]);
 
 /* ConstructCommitPayload (consensus/signature/signature.go:12-24); returns 40/48 */
 int hbls_construct_commit_payload(uint64_t block_num, const uint8_t hash32[32],

@@ -562,3 +562,22 @@ def test_verify_rf_parity(core, capi, keys16):
                 core.set_coop_threshold(-1)
     finally:
         core._lib.hbls_set_verify_rf(-1)
+
+
+def test_msm_committee_parity(core, capi, committee4096):
+    """hbls_msm_g1_committee: Pippenger core over the resident validated
+    table — same result as the serialized-points entry and the python-int
+    expectation (points are pk_i = sk_i*G)."""
+    import random
+    sks, pks, gc = committee4096
+    n = 4096
+    rng = random.Random(77)
+    scalars = [rng.randrange(pr.R) for _ in range(n)]
+    sc = b"".join(pr.fr_serialize(s) for s in scalars)
+    got = gc.msm(sc)
+    assert got == core.msm_g1(pks, sc, n)
+    sk_ints = [pr.synth_sk(i) for i in range(n)]
+    acc = sum(s * k for s, k in zip(scalars, sk_ints)) % pr.R
+    assert got == capi.pk_from_sk(pr.fr_serialize(acc))
+    with pytest.raises(ValueError):
+        gc.msm(sc[:-32])
