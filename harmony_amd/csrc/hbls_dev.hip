@@ -2459,34 +2459,64 @@ __global__ void __launch_bounds__(64) k_msm_buckets(
     if (threadIdx.x == 0) buckets[w * 255 + b - 1] = red[0];
 }
 
-__global__ void __launch_bounds__(64) k_msm_wreduce(const g1_t *buckets, g1_t *wsums) {
+/* window reduction, parallelized (the serial 510-add suffix sum per window
+ * was a 23 ms tail — profiles/r02_data/r2k_msm_kernels.txt).  Chunked:
+ *   S_w = sum_b b*B_b = sum_c [ P_c + (lo_c - 1)*T_c ]
+ * with P_c the chunk-local weighted suffix sum and T_c the chunk total;
+ * (lo_c - 1) = 16c, and sum_c 16c*T_c = 16 * sum of T-suffix sums. */
+__global__ void __launch_bounds__(64) k_msm_wreduce(const g1_t *buckets,
+                                                    g1_t *chunkP, g1_t *chunkT) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;   /* 32 windows x 16 chunks */
+    if (t >= 32 * 16) return;
+    int w = t / 16, c = t % 16;
+    int lo = c * 16 + 1;
+    int hi = lo + 16 < 256 ? lo + 16 : 256;
+    g1_t run, wsum;
+    g1_set_inf(run);
+    g1_set_inf(wsum);
+    for (int b = hi - 1; b >= lo; b--) {
+        g1_t t1;
+        g1_add(t1, run, buckets[w * 255 + b - 1]);
+        run = t1;
+        g1_add(t1, wsum, run);
+        wsum = t1;
+    }
+    chunkP[t] = wsum;
+    chunkT[t] = run;
+}
+__global__ void __launch_bounds__(64) k_msm_wcombine(const g1_t *chunkP,
+                                                     const g1_t *chunkT, g1_t *wsums) {
     int w = blockIdx.x * blockDim.x + threadIdx.x;
     if (w >= 32) return;
-    g1_t running, sum;
-    g1_set_inf(running);
-    g1_set_inf(sum);
-    for (int b = 254; b >= 0; b--) {
-        g1_t t;
-        g1_add(t, running, buckets[w * 255 + b]);
-        running = t;
-        g1_add(t, sum, running);
-        sum = t;
+    g1_t acc, t;
+    g1_set_inf(acc);
+    for (int c = 0; c < 16; c++) { g1_add(t, acc, chunkP[w * 16 + c]); acc = t; }
+    g1_t run, ssum;
+    g1_set_inf(run);
+    g1_set_inf(ssum);
+    for (int c = 15; c >= 1; c--) {
+        g1_add(t, run, chunkT[w * 16 + c]);
+        run = t;
+        g1_add(t, ssum, run);
+        ssum = t;
     }
-    wsums[w] = sum;
+    for (int d = 0; d < 4; d++) { g1_dbl(t, ssum); ssum = t; }   /* x16 */
+    g1_add(t, acc, ssum);
+    wsums[w] = t;
 }
-
-__global__ void k_msm_combine(const g1_t *wsums, uint8_t *out48) {
-    g1_t acc = wsums[31];
-    for (int w = 30; w >= 0; w--) {
-        for (int d = 0; d < 8; d++) {
-            g1_t t;
-            g1_dbl(t, acc);
-            acc = t;
-        }
-        g1_t t;
-        g1_add(t, acc, wsums[w]);
-        acc = t;
-    }
+/* Horner combine, parallelized: each window shifts by its own 8w doublings
+ * (longest chain 248, all windows concurrent), then one 32-add fold. */
+__global__ void __launch_bounds__(64) k_msm_shift(const g1_t *wsums, g1_t *shifted) {
+    int w = blockIdx.x * blockDim.x + threadIdx.x;
+    if (w >= 32) return;
+    g1_t acc = wsums[w], t;
+    for (int d = 0; d < 8 * w; d++) { g1_dbl(t, acc); acc = t; }
+    shifted[w] = acc;
+}
+__global__ void k_msm_final(const g1_t *shifted, uint8_t *out48) {
+    g1_t acc, t;
+    g1_set_inf(acc);
+    for (int w = 0; w < 32; w++) { g1_add(t, acc, shifted[w]); acc = t; }
     g1_serialize(out48, acc);
 }
 
@@ -2524,13 +2554,19 @@ static int msm_core(const g1aff_t *d_pts, const uint8_t *d_digits, size_t n,
                        dcnt.as<uint32_t>(), doff.as<uint32_t>(), dcur.as<uint32_t>());
     hipLaunchKernelGGL(k_msm_scatter, dim3(nb_sc), dim3(256), 0, 0,
                        d_digits, (int)n, dcur.as<uint32_t>(), dlist.as<uint32_t>());
+    DevBuf dcp(32 * 16 * sizeof(g1_t)), dct(32 * 16 * sizeof(g1_t)), dsh(32 * sizeof(g1_t));
+    if (dcp.err || dct.err || dsh.err) return HBLS_ERR;
     hipLaunchKernelGGL(k_msm_buckets, dim3(32 * 255), dim3(64), 0, 0,
                        d_pts, dlist.as<uint32_t>(), doff.as<uint32_t>(),
                        (int)n, dbuck.as<g1_t>());
-    hipLaunchKernelGGL(k_msm_wreduce, dim3(1), dim3(64), 0, 0,
-                       dbuck.as<g1_t>(), dws.as<g1_t>());
-    hipLaunchKernelGGL(k_msm_combine, dim3(1), dim3(1), 0, 0,
-                       dws.as<g1_t>(), d_out48);
+    hipLaunchKernelGGL(k_msm_wreduce, dim3(8), dim3(64), 0, 0,
+                       dbuck.as<g1_t>(), dcp.as<g1_t>(), dct.as<g1_t>());
+    hipLaunchKernelGGL(k_msm_wcombine, dim3(1), dim3(64), 0, 0,
+                       dcp.as<g1_t>(), dct.as<g1_t>(), dws.as<g1_t>());
+    hipLaunchKernelGGL(k_msm_shift, dim3(1), dim3(64), 0, 0,
+                       dws.as<g1_t>(), dsh.as<g1_t>());
+    hipLaunchKernelGGL(k_msm_final, dim3(1), dim3(1), 0, 0,
+                       dsh.as<g1_t>(), d_out48);
     HIP_OK(hipGetLastError());
     /* keep the intermediate buffers alive until the chain completes (hipFree
      * in ~DevBuf synchronizes, but be explicit about the dependency) */
@@ -2569,36 +2605,20 @@ extern "C" int hbls_msm_g1(const uint8_t *points48, const uint8_t *scalars32, si
     if (rc != HBLS_OK) return rc;
     if (n == 0) return HBLS_ERR_BADINPUT;
     DevBuf dp(n * 48), ds(n * 32), dpts(n * sizeof(g1aff_t)), ddig(n * 32);
-    DevBuf dcnt(32 * 256 * 4), doff(32 * 257 * 4), dcur(32 * 256 * 4), dlist(n * 32 * 4);
-    DevBuf dbuck(32 * 255 * sizeof(g1_t)), dws(32 * sizeof(g1_t)), dok(4), dout(48);
-    if (dp.err || ds.err || dpts.err || ddig.err || dcnt.err || doff.err ||
-        dcur.err || dlist.err || dbuck.err || dws.err ||
-        dok.err || dout.err) return HBLS_ERR;
+    DevBuf dok(4), dout(48);
+    if (dp.err || ds.err || dpts.err || ddig.err || dok.err || dout.err)
+        return HBLS_ERR;
     HIP_OK(hipMemcpy(dp.p, points48, n * 48, hipMemcpyHostToDevice));
     HIP_OK(hipMemcpy(ds.p, scalars32, n * 32, hipMemcpyHostToDevice));
     int32_t one = 1;
     HIP_OK(hipMemcpy(dok.p, &one, 4, hipMemcpyHostToDevice));
-    HIP_OK(hipMemset(dcnt.p, 0, 32 * 256 * 4));
     Timer tm;
     hipLaunchKernelGGL(k_msm_prep, dim3((uint32_t)((n + 63) / 64)), dim3(64), 0, 0,
                        dp.as<uint8_t>(), ds.as<uint8_t>(), (int)n,
                        dpts.as<g1aff_t>(), ddig.as<uint8_t>(), dok.as<int32_t>());
-    int nb_sc = (int)(((size_t)32 * n + 255) / 256);
-    if (nb_sc > 4096) nb_sc = 4096;
-    hipLaunchKernelGGL(k_msm_hist, dim3(nb_sc), dim3(256), 0, 0,
-                       ddig.as<uint8_t>(), (int)n, dcnt.as<uint32_t>());
-    hipLaunchKernelGGL(k_msm_prefix, dim3(1), dim3(64), 0, 0,
-                       dcnt.as<uint32_t>(), doff.as<uint32_t>(), dcur.as<uint32_t>());
-    hipLaunchKernelGGL(k_msm_scatter, dim3(nb_sc), dim3(256), 0, 0,
-                       ddig.as<uint8_t>(), (int)n, dcur.as<uint32_t>(), dlist.as<uint32_t>());
-    hipLaunchKernelGGL(k_msm_buckets, dim3(32 * 255), dim3(64), 0, 0,
-                       dpts.as<g1aff_t>(), dlist.as<uint32_t>(), doff.as<uint32_t>(),
-                       (int)n, dbuck.as<g1_t>());
-    hipLaunchKernelGGL(k_msm_wreduce, dim3(1), dim3(64), 0, 0,
-                       dbuck.as<g1_t>(), dws.as<g1_t>());
-    hipLaunchKernelGGL(k_msm_combine, dim3(1), dim3(1), 0, 0,
-                       dws.as<g1_t>(), dout.as<uint8_t>());
+    rc = msm_core(dpts.as<g1aff_t>(), ddig.as<uint8_t>(), n, dout.as<uint8_t>());
     tm.stop_and_store();
+    if (rc != HBLS_OK) return rc;
     HIP_OK(hipGetLastError());
     int32_t ok;
     HIP_OK(hipMemcpy(&ok, dok.p, 4, hipMemcpyDeviceToHost));
