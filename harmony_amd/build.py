@@ -18,6 +18,18 @@ CMD = [
 # instrumented twin (fp_mul call counter; loaded explicitly by experiments)
 OUT_COUNT = os.path.join(DIR, "libhbls_count.so")
 CMD_COUNT = CMD[:-1] + [OUT_COUNT, "-DHBLS_COUNT_MULS"]
+# compact-code twin: fp2 ops call the register-ABI CIOS instead of inlining it
+OUT_CS = os.path.join(DIR, "libhbls_cs.so")
+CMD_CS = CMD[:-1] + [OUT_CS, "-DHBLS_FP2_RS"]
+
+
+def build_cs(force=False):
+    if not force and os.path.exists(OUT_CS) and \
+            os.path.getmtime(OUT_CS) > os.path.getmtime(SRC):
+        return OUT_CS
+    print("+", " ".join(CMD_CS), flush=True)
+    subprocess.check_call(CMD_CS)
+    return OUT_CS
 
 
 def build_count(force=False):

@@ -8,7 +8,7 @@ import ctypes
 import os
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
-_SO = os.path.join(_DIR, "libhbls.so")
+_SO = os.environ.get("HBLS_SO", os.path.join(_DIR, "libhbls.so"))
 
 HBLS_OK = 1
 HBLS_FALSE = 0

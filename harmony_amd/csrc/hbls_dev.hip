@@ -281,6 +281,24 @@ DEVN bool fp_is_odd(const fp_t &x) {
     return raw[0] & 1;
 }
 
+
+/* register-ABI multiplier: 12 scalar u64 args + 12-dword struct return ride
+ * entirely in VGPRs (zero scratch at the call boundary — measured, see
+ * DESIGN.md §4b).  Shared by the HBLS_FP2_RS compact-code experiment and
+ * the opt-in rf build. */
+DEVN fp_t fp_mul_rs(uint64_t x0, uint64_t x1, uint64_t x2, uint64_t x3,
+                    uint64_t x4, uint64_t x5,
+                    uint64_t y0, uint64_t y1, uint64_t y2, uint64_t y3,
+                    uint64_t y4, uint64_t y5) {
+    fp_t x, y, r;
+    x.l[0] = x0; x.l[1] = x1; x.l[2] = x2; x.l[3] = x3; x.l[4] = x4; x.l[5] = x5;
+    y.l[0] = y0; y.l[1] = y1; y.l[2] = y2; y.l[3] = y3; y.l[4] = y4; y.l[5] = y5;
+    FP_MUL_BODY(r, x, y);
+    return r;
+}
+#define RFM(r, x, y) (r) = fp_mul_rs((x).l[0], (x).l[1], (x).l[2], (x).l[3], (x).l[4], (x).l[5], \
+                                     (y).l[0], (y).l[1], (y).l[2], (y).l[3], (y).l[4], (y).l[5])
+
 /* ================================================================ Fp2 */
 DEV bool fp2_is_zero(const fp2_t &x) { return fp_is_zero(x.a) && fp_is_zero(x.b); }
 DEV bool fp2_eq(const fp2_t &x, const fp2_t &y) { return fp_eq(x.a, y.a) && fp_eq(x.b, y.b); }
@@ -290,6 +308,22 @@ DEV void fp2_neg(fp2_t &r, const fp2_t &x) { fp_neg(r.a, x.a); fp_neg(r.b, x.b);
 DEV void fp2_conj(fp2_t &r, const fp2_t &x) { r.a = x.a; fp_neg(r.b, x.b); }
 DEV void fp2_dbl(fp2_t &r, const fp2_t &x) { fp2_add(r, x, x); }
 DEVN void fp2_mul(fp2_t &r, const fp2_t &x, const fp2_t &y) {
+#ifdef HBLS_FP2_RS
+    /* compact-code variant: ~400-instr body + calls into the shared
+     * register-ABI CIOS, instead of three inlined ~1.4k-instr CIOS copies
+     * (the inlined form makes fp2_mul a 5k-instr function and the Miller
+     * call set ~240 KB vs the 32 KB I-cache) */
+    fp_t ac, bd, ab, cd, t;
+    RFM(ac, x.a, y.a);
+    RFM(bd, x.b, y.b);
+    fp_add(ab, x.a, x.b);
+    fp_add(cd, y.a, y.b);
+    RFM(t, ab, cd);
+    fp_sub(t, t, ac);
+    fp_sub(t, t, bd);
+    fp_sub(r.a, ac, bd);
+    r.b = t;
+#else
     fp_t ac, bd, ab, cd, t;
     fp_mul_inl(ac, x.a, y.a);
     fp_mul_inl(bd, x.b, y.b);
@@ -300,14 +334,24 @@ DEVN void fp2_mul(fp2_t &r, const fp2_t &x, const fp2_t &y) {
     fp_sub(t, t, bd);
     fp_sub(r.a, ac, bd);
     r.b = t;
+#endif
 }
 DEVN void fp2_sqr(fp2_t &r, const fp2_t &x) {
+#ifdef HBLS_FP2_RS
+    fp_t s, d, m;
+    fp_add(s, x.a, x.b);
+    fp_sub(d, x.a, x.b);
+    RFM(m, x.a, x.b);
+    RFM(r.a, s, d);
+    fp_dbl(r.b, m);
+#else
     fp_t s, d, m;
     fp_add(s, x.a, x.b);
     fp_sub(d, x.a, x.b);
     fp_mul_inl(m, x.a, x.b);
     fp_mul_inl(r.a, s, d);
     fp_dbl(r.b, m);
+#endif
 }
 DEV void fp2_mul_fp(fp2_t &r, const fp2_t &x, const fp_t &s) {
     fp_mul(r.a, x.a, s);
