@@ -215,7 +215,7 @@ def run_stream(args):
     from harmony_amd.stream import MultiStreamVerifier, StreamVerifier
     from oracle import capi, pyref as pr
     n = 256                     # mainnet per-shard committee is 250 keys
-    R = 16                      # rounds in flight
+    R = int(os.environ.get("HBLS_STREAM_ROUNDS", "16"))   # rounds in flight
     sks = [pr.fr_serialize(pr.synth_sk(i)) for i in range(n)]
     pks = core.batch_pk_from_sk(b"".join(sks), n)
     blob_len = 512
