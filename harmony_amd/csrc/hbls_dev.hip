@@ -2056,11 +2056,15 @@ static int rf_mode(void) {
 }
 extern "C" void hbls_set_verify_rf(int mode) { g_rf_override = mode; }
 /* batched Montgomery-inversion affine conversion ahead of the scalar verify
- * kernel (default ON; HBLS_NO_BATCH_AFFINE=1 restores the round-1 jacobian
- * path for A/Bs).  Only the default kernel mode uses it. */
+ * kernel.  MEASURED WALL-NEUTRAL (+1.7 ms on a 197 ms launch at batch
+ * 131072, profiles/r02_data/r2o_affine_*.json) despite removing ~5% of the
+ * multiplies — the fifth datapoint confirming the pairing kernel is bound
+ * by scratch traffic, not arithmetic (with r1's fused-Miller neutrality
+ * and the four r2 falsifications).  Default OFF; HBLS_BATCH_AFFINE=1
+ * enables for A/Bs. */
 static int use_batch_affine(void) {
     static int m = -1;
-    if (m < 0) { const char *e = getenv("HBLS_NO_BATCH_AFFINE"); m = (e && atoi(e)) ? 0 : 1; }
+    if (m < 0) { const char *e = getenv("HBLS_BATCH_AFFINE"); m = (e && atoi(e)) ? 1 : 0; }
     return m;
 }
 #ifdef HBLS_RF
