@@ -1175,6 +1175,195 @@ DEVN void final_exp(fp12_t &r, const fp12_t &f_in) {
     fp12_mul(r, d, t);
 }
 /* verify core: pub (G1 jac), sig (G2 affine, pre-checked), hm (G2 jac) */
+/* ---- fp6-granularity called layer (traffic experiment, HBLS_VERIFY_RF=5):
+ * the verify kernel's scratch traffic is ~proportional to the number of
+ * by-ref field-op calls (each fp2-level call round-trips ~290 B of
+ * operands).  Coarsening the called unit to fp6 (864 B per fp6-mul vs
+ * ~1.7 KB for its six fp2 calls) with ALL internal multiplies through the
+ * zero-traffic register-ABI CIOS should cut the fp6-shaped ~60% of the
+ * traffic nearly in half — IF the traffic model is right. */
+DEV void fp2_mul_ri(fp2_t &r, const fp2_t &x, const fp2_t &y) {
+    fp_t ac, bd, ab, cd, t;
+    RFM(ac, x.a, y.a);
+    RFM(bd, x.b, y.b);
+    fp_add(ab, x.a, x.b);
+    fp_add(cd, y.a, y.b);
+    RFM(t, ab, cd);
+    fp_sub(t, t, ac);
+    fp_sub(t, t, bd);
+    fp_sub(r.a, ac, bd);
+    r.b = t;
+}
+DEVN void fp6_mul_rs6(fp6_t &r, const fp6_t &x, const fp6_t &y) {
+    fp2_t t0, t1, t2, s0, s1, tt, r0, r1;
+    fp2_mul_ri(t0, x.c0, y.c0);
+    fp2_mul_ri(t1, x.c1, y.c1);
+    fp2_mul_ri(t2, x.c2, y.c2);
+    fp2_add(s0, x.c1, x.c2);
+    fp2_add(s1, y.c1, y.c2);
+    fp2_mul_ri(tt, s0, s1);
+    fp2_sub(tt, tt, t1);
+    fp2_sub(tt, tt, t2);
+    fp2_mul_xi(tt, tt);
+    fp2_add(r0, t0, tt);
+    fp2_add(s0, x.c0, x.c1);
+    fp2_add(s1, y.c0, y.c1);
+    fp2_mul_ri(tt, s0, s1);
+    fp2_sub(tt, tt, t0);
+    fp2_sub(tt, tt, t1);
+    fp2_t xt2;
+    fp2_mul_xi(xt2, t2);
+    fp2_add(r1, tt, xt2);
+    fp2_add(s0, x.c0, x.c2);
+    fp2_add(s1, y.c0, y.c2);
+    fp2_mul_ri(tt, s0, s1);
+    fp2_sub(tt, tt, t0);
+    fp2_sub(tt, tt, t2);
+    fp2_add(r.c2, tt, t1);
+    r.c0 = r0; r.c1 = r1;
+}
+DEVN void fp12_mul_6(fp12_t &r, const fp12_t &x, const fp12_t &y) {
+    fp6_t t0, t1, s0, s1, tt, vt1;
+    fp6_mul_rs6(t0, x.c0, y.c0);
+    fp6_mul_rs6(t1, x.c1, y.c1);
+    fp6_add(s0, x.c0, x.c1);
+    fp6_add(s1, y.c0, y.c1);
+    fp6_mul_rs6(tt, s0, s1);
+    fp6_sub(tt, tt, t0);
+    fp6_sub(tt, tt, t1);
+    fp6_mul_v(vt1, t1);
+    fp6_add(r.c0, t0, vt1);
+    r.c1 = tt;
+}
+DEVN void fp12_sqr_6(fp12_t &r, const fp12_t &x) {
+    fp6_t ab, apb, avb, t0, vab;
+    fp6_mul_rs6(ab, x.c0, x.c1);
+    fp6_add(apb, x.c0, x.c1);
+    fp6_mul_v(avb, x.c1);
+    fp6_add(avb, x.c0, avb);
+    fp6_mul_rs6(t0, apb, avb);
+    fp6_sub(t0, t0, ab);
+    fp6_mul_v(vab, ab);
+    fp6_sub(r.c0, t0, vab);
+    fp6_add(r.c1, ab, ab);
+}
+DEVN void fp12_mul_line_6(fp12_t &f, const fp2_t &c0, const fp2_t &c3, const fp2_t &c5) {
+    fp6_t t0, t1, l01, tt, vt1, fs, l;
+    fp2_mul(t0.c0, f.c0.c0, c0);
+    fp2_mul(t0.c1, f.c0.c1, c0);
+    fp2_mul(t0.c2, f.c0.c2, c0);
+    {
+        const fp2_t &a0 = f.c1.c0, &a1 = f.c1.c1, &a2 = f.c1.c2;
+        fp2_t p1, p2, q;
+        fp2_mul(p1, a1, c5);
+        fp2_mul(p2, a2, c3);
+        fp2_add(q, p1, p2);
+        fp2_mul_xi(t1.c0, q);
+        fp2_mul(p1, a0, c3);
+        fp2_mul(p2, a2, c5);
+        fp2_mul_xi(p2, p2);
+        fp2_add(t1.c1, p1, p2);
+        fp2_mul(p1, a0, c5);
+        fp2_mul(p2, a1, c3);
+        fp2_add(t1.c2, p1, p2);
+    }
+    fp6_add(fs, f.c0, f.c1);
+    l.c0 = c0; l.c1 = c3; l.c2 = c5;
+    fp6_mul_rs6(l01, fs, l);
+    fp6_sub(tt, l01, t0);
+    fp6_sub(tt, tt, t1);
+    fp6_mul_v(vt1, t1);
+    fp6_add(f.c0, t0, vt1);
+    f.c1 = tt;
+}
+DEVN void miller_loop2_6(fp12_t &f, const g2aff_t &Q1, const g1aff_t &P1,
+                         const g2aff_t &Q2, const g1aff_t &P2) {
+    g2_t T1, T2;
+    g2_from_affine(T1, Q1);
+    g2_from_affine(T2, Q2);
+    fp2_t c0, c3, c5;
+    fp12_one(f);
+    for (int bit = 62; bit >= 0; bit--) {
+        fp12_sqr_6(f, f);
+        ml_dbl_step(T1, P1, c0, c3, c5);
+        fp12_mul_line_6(f, c0, c3, c5);
+        ml_dbl_step(T2, P2, c0, c3, c5);
+        fp12_mul_line_6(f, c0, c3, c5);
+        if ((BLS_U >> bit) & 1) {
+            ml_add_step(T1, Q1, P1, c0, c3, c5);
+            fp12_mul_line_6(f, c0, c3, c5);
+            ml_add_step(T2, Q2, P2, c0, c3, c5);
+            fp12_mul_line_6(f, c0, c3, c5);
+        }
+    }
+}
+DEVN void fp12_pow_u_6(fp12_t &r, const fp12_t &x) {
+    fp12_t acc = x;
+    for (int bit = 62; bit >= 0; bit--) {
+        fp12_cyc_sqr(acc, acc);
+        if ((BLS_U >> bit) & 1) fp12_mul_6(acc, acc, x);
+    }
+    r = acc;
+}
+DEVN void final_exp_6(fp12_t &r, const fp12_t &f_in) {
+    fp12_t f, t, inv;
+    fp12_conj(t, f_in);
+    fp12_inv(inv, f_in);
+    fp12_mul_6(f, t, inv);
+    fp12_frob2(t, f);
+    fp12_mul_6(f, t, f);
+    fp12_t a, b, c, d, u1, u2;
+    fp12_pow_u_6(u1, f);
+    fp12_mul_6(a, u1, f);
+    fp12_conj(a, a);
+    fp12_pow_u_6(u1, a);
+    fp12_mul_6(b, u1, a);
+    fp12_conj(b, b);
+    fp12_pow_u_6(u1, b);
+    fp12_conj(u1, u1);
+    fp12_frob(u2, b);
+    fp12_mul_6(c, u1, u2);
+    fp12_pow_u_6(u1, c);
+    fp12_pow_u_6(u1, u1);
+    fp12_frob2(u2, c);
+    fp12_mul_6(d, u1, u2);
+    fp12_conj(u1, c);
+    fp12_mul_6(d, d, u1);
+    fp12_sqr_6(t, f);
+    fp12_mul_6(t, t, f);
+    fp12_mul_6(r, d, t);
+}
+DEVN int verify_pairing_6(const g1_t &pub, const g2_t &hm, const g2aff_t &sig_aff,
+                          bool sig_inf) {
+    bool pub_inf = g1_is_inf(pub);
+    if (pub_inf && sig_inf) return 1;
+    if (pub_inf || sig_inf) return 0;
+    g1aff_t pa, ba;
+    g2aff_t ha;
+    g1_to_affine(pa, pub);
+    { g1_t base, nb;
+      fp_load(base.x, BLS_G1_X); fp_load(base.y, BLS_G1_Y); fp_one(base.z);
+      g1_neg(nb, base);
+      ba.x = nb.x; ba.y = nb.y; }
+    g2_t hmc = hm;
+    g2_to_affine(ha, hmc);
+    fp12_t f;
+    miller_loop2_6(f, ha, pa, sig_aff, ba);
+    fp12_conj(f, f);
+    final_exp_6(f, f);
+    return fp12_is_one(f) ? 1 : 0;
+}
+__global__ void k_verify_6(const g1_t *aggpubs, const g2_t *hms, const g2aff_t *sigs,
+                           const int32_t *sig_flags, const int32_t *hm_ok,
+                           int32_t *results, int batch) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= batch) return;
+    if (!hm_ok[i] || sig_flags[i] == 0) { results[i] = HBLS_ERR_BADINPUT; return; }
+    g2aff_t dummy;
+    bool sig_inf = sig_flags[i] == 2;
+    results[i] = verify_pairing_6(aggpubs[i], hms[i], sig_inf ? dummy : sigs[i], sig_inf);
+}
+
 /* affine-input pairing: identical math to verify_pairing below, with the
  * per-item fp_inv/fp2_inv conversions hoisted into the batched Montgomery-
  * inversion kernels (k_g1_batch_affine / k_g2_batch_affine) — ~1.1k of the
@@ -2079,6 +2268,7 @@ static int use_batch_affine(void) {
     switch (rf_mode()) { \
     HBLS_RF_CASES(nb, __VA_ARGS__) \
     case 4: hipLaunchKernelGGL(k_verify_2p, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
+    case 5: hipLaunchKernelGGL(k_verify_6, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
     default: hipLaunchKernelGGL(k_verify, dim3(nb), dim3(64), 0, 0, __VA_ARGS__); break; \
     } \
 } while (0)

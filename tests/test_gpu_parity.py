@@ -547,7 +547,7 @@ def test_verify_rf_parity(core, capi, keys16):
     bad = bytearray(bm)
     bad[0] ^= 1
     core._lib.hbls_set_verify_rf.argtypes = [__import__("ctypes").c_int]
-    modes = (0, 4) if not core._lib.hbls_has_rf() else (0, 1, 2, 3, 4)
+    modes = (0, 4, 5) if not core._lib.hbls_has_rf() else (0, 1, 2, 3, 4, 5)
     try:
         for mode in modes:
             core._lib.hbls_set_verify_rf(mode)
