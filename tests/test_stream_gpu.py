@@ -153,3 +153,29 @@ def test_device_stream_context(oracle_lib):
         st.process([0], [R + 5], sig_of[(0, 0)])
     with pytest.raises(ValueError):
         st.check([R])
+
+
+def test_stream_ragged_committee(oracle_lib):
+    """n=33: the device bitmap is 32-bit words with a ragged tail — dedup
+    across the word boundary (keys 31/32) and byte-exact export."""
+    from harmony_amd import core
+    n = 33
+    sks = [pr.fr_serialize(pr.synth_sk(i)) for i in range(n)]
+    pks = core.batch_pk_from_sk(b"".join(sks), n)
+    st = core.Stream(core.Committee(pks, n), 2)
+    payloads = [pr.construct_commit_payload(r, pr.synth_msg(300 + r), r)
+                for r in range(2)]
+    st.set_rounds([0, 1], b"".join(payloads), len(payloads[0]))
+    votes = [(r, i) for r in range(2) for i in (31, 32, 0, 32)]   # dups incl. key 32
+    sigs = b"".join(oracle_lib.sign_hash(sks[i], payloads[r]) for r, i in votes)
+    res = st.process([v[1] for v in votes], [v[0] for v in votes], sigs)
+    for r in range(2):
+        sub = res[4 * r:4 * r + 4]
+        assert sorted(sub) == [1, 1, 1, 2], res      # one dup of key 32 per round
+    assert st.check([0, 1]) == [True, True]
+    oc = oracle_lib.Committee(pks, n)
+    for r in range(2):
+        bm, agg = st.get(r)
+        assert len(bm) == 5
+        assert bm == bytes([0x01, 0x00, 0x00, 0x80, 0x01])   # keys 0, 31, 32
+        assert oc.agg_verify(bm, agg, payloads[r]) is True
