@@ -78,6 +78,10 @@ def _load():
     lib.hbls_stream_check.argtypes = [
         ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32), ctypes.c_int,
         ctypes.POINTER(ctypes.c_int32)]
+    lib.hbls_stream_check_submit.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32), ctypes.c_int]
+    lib.hbls_stream_check_poll.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_int32)]
     lib.hbls_stream_get.argtypes = [ctypes.c_void_p, ctypes.c_uint32,
                                     ctypes.c_char_p, ctypes.c_char_p]
     lib.hbls_fpmul_bench_waves.restype = ctypes.c_double
@@ -357,6 +361,24 @@ class Stream:
         ok = (ctypes.c_int32 * k)()
         _check(_lib.hbls_stream_check(self._h, arr, k, ok), "stream_check")
         return [bool(v == 1) for v in ok]
+
+    def check_submit(self, slots):
+        """async check: snapshots the rounds and runs the pairing chain on a
+        side stream (overlaps later process() ticks); one in flight."""
+        k = len(slots)
+        arr = (ctypes.c_uint32 * k)(*slots)
+        _check(_lib.hbls_stream_check_submit(self._h, arr, k), "stream_check_submit")
+        self._pending_check = list(slots)
+
+    def check_poll(self):
+        """waits for the in-flight check; returns {slot: bool} (empty if none)"""
+        slots = getattr(self, "_pending_check", None)
+        if not slots:
+            return {}
+        ok = (ctypes.c_int32 * len(slots))()
+        k = _check(_lib.hbls_stream_check_poll(self._h, ok), "stream_check_poll")
+        self._pending_check = None
+        return {s: bool(ok[i] == 1) for i, s in enumerate(slots[:k])}
 
     def get(self, slot: int):
         """returns (bitmap bytes, serialized aggregate sig 96B)"""

@@ -1834,28 +1834,29 @@ k_mask_aggregate_w(const g1aff_t *wtab, const uint8_t *winf, int n,
 static inline void launch_mask_aggregate(const g1aff_t *table, int n,
         const uint8_t *bm, int bm_stride, const g1_t *full_sum,
         g1_t *out, int batch,
-        const g1aff_t *wtab = nullptr, const uint8_t *winf = nullptr) {
+        const g1aff_t *wtab = nullptr, const uint8_t *winf = nullptr,
+        hipStream_t stream = 0) {
     int blocks = (batch + MASK_SUBS - 1) / MASK_SUBS;
     if (wtab != nullptr) {
         /* small bitmaps leave 64 lanes with so few byte-adds that the
          * reduction tree dominates: use 16 lanes below 2048 bytes */
         if (bm_stride >= 2048)
             hipLaunchKernelGGL((k_mask_aggregate_w<64>), dim3(blocks),
-                               dim3(64 * MASK_SUBS), 0, 0,
+                               dim3(64 * MASK_SUBS), 0, stream,
                                wtab, winf, n, bm, bm_stride, full_sum, out, batch);
         else
             hipLaunchKernelGGL((k_mask_aggregate_w<16>), dim3(blocks),
-                               dim3(16 * MASK_SUBS), 0, 0,
+                               dim3(16 * MASK_SUBS), 0, stream,
                                wtab, winf, n, bm, bm_stride, full_sum, out, batch);
         return;
     }
     if (n >= 16384)
         hipLaunchKernelGGL((k_mask_aggregate<64>), dim3(blocks),
-                           dim3(64 * MASK_SUBS), 0, 0,
+                           dim3(64 * MASK_SUBS), 0, stream,
                            table, n, bm, bm_stride, full_sum, out, batch);
     else
         hipLaunchKernelGGL((k_mask_aggregate<MASK_LANES>), dim3(blocks),
-                           dim3(MASK_BLOCK), 0, 0,
+                           dim3(MASK_BLOCK), 0, stream,
                            table, n, bm, bm_stride, full_sum, out, batch);
 }
 
@@ -3724,6 +3725,19 @@ struct hbls_stream {
     int32_t *d_hm_ok;
     uint32_t *d_bitmap;   /* max_rounds x nwords, little-endian bit order */
     g2_t *d_agg;          /* per-round aggregate signature (jacobian) */
+    /* async check lane: the periodic window checks (a latency-bound
+     * 1-block pairing for <=16 rounds) run on their own HIP stream against
+     * SNAPSHOTS of the rounds' state, overlapping the next ticks */
+    hipStream_t check_stream;
+    uint8_t *ck_bm;       /* snapshots + scratch, sized max_rounds */
+    g2_t *ck_hm;
+    int32_t *ck_hok;
+    g2aff_t *ck_saff;
+    int32_t *ck_sflags;
+    g1_t *ck_agg;
+    int32_t *ck_res;
+    uint32_t *ck_slots;
+    int ck_pending;       /* rounds in the in-flight check (0 = none) */
 };
 
 __global__ void k_stream_reset(uint32_t *bitmap, g2_t *agg, const uint32_t *slots,
@@ -3860,16 +3874,42 @@ extern "C" hbls_stream *hbls_stream_create(const hbls_committee_t *c, int max_ro
     (void)hipMemcpy(dslots.p, all.data(), R * 4, hipMemcpyHostToDevice);
     hipLaunchKernelGGL(k_stream_reset, dim3((uint32_t)R), dim3(64), 0, 0,
                        s->d_bitmap, s->d_agg, dslots.as<uint32_t>(), (int)R, s->nwords);
+    size_t bm = (c->n + 7) / 8;
+    if (hipStreamCreate(&s->check_stream) != hipSuccess ||
+        hipMalloc(&s->ck_bm, R * bm) != hipSuccess ||
+        hipMalloc(&s->ck_hm, R * sizeof(g2_t)) != hipSuccess ||
+        hipMalloc(&s->ck_hok, R * 4) != hipSuccess ||
+        hipMalloc(&s->ck_saff, R * sizeof(g2aff_t)) != hipSuccess ||
+        hipMalloc(&s->ck_sflags, R * 4) != hipSuccess ||
+        hipMalloc(&s->ck_agg, R * sizeof(g1_t)) != hipSuccess ||
+        hipMalloc(&s->ck_res, R * 4) != hipSuccess ||
+        hipMalloc(&s->ck_slots, R * 4) != hipSuccess) {
+        hbls_stream_free(s);
+        return nullptr;
+    }
+    s->ck_pending = 0;
     if (hipDeviceSynchronize() != hipSuccess) { hbls_stream_free(s); return nullptr; }
     return s;
 }
 
 extern "C" void hbls_stream_free(hbls_stream *s) {
     if (!s) return;
+    if (s->check_stream) {
+        (void)hipStreamSynchronize(s->check_stream);
+        (void)hipStreamDestroy(s->check_stream);
+    }
     if (s->d_hm) (void)hipFree(s->d_hm);
     if (s->d_hm_ok) (void)hipFree(s->d_hm_ok);
     if (s->d_bitmap) (void)hipFree(s->d_bitmap);
     if (s->d_agg) (void)hipFree(s->d_agg);
+    if (s->ck_bm) (void)hipFree(s->ck_bm);
+    if (s->ck_hm) (void)hipFree(s->ck_hm);
+    if (s->ck_hok) (void)hipFree(s->ck_hok);
+    if (s->ck_saff) (void)hipFree(s->ck_saff);
+    if (s->ck_sflags) (void)hipFree(s->ck_sflags);
+    if (s->ck_agg) (void)hipFree(s->ck_agg);
+    if (s->ck_res) (void)hipFree(s->ck_res);
+    if (s->ck_slots) (void)hipFree(s->ck_slots);
     delete s;
 }
 
@@ -4011,6 +4051,53 @@ extern "C" int hbls_stream_check(hbls_stream *s, const uint32_t *slots, int k,
     HIP_OK(hipGetLastError());
     HIP_OK(hipMemcpy(ok, dres.p, k * 4, hipMemcpyDeviceToHost));
     return HBLS_OK;
+}
+
+/* async window check: snapshot the rounds' state on the DEFAULT stream
+ * (ordered after the ticks that built it), then run the latency-bound
+ * mask+pairing chain on a dedicated HIP stream so it overlaps subsequent
+ * ticks.  Poll with hbls_stream_check_poll; at most one check in flight. */
+extern "C" int hbls_stream_check_submit(hbls_stream *s, const uint32_t *slots, int k) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    if (k <= 0 || k > s->max_rounds || s->ck_pending) return HBLS_ERR_BADINPUT;
+    for (int i = 0; i < k; i++)
+        if (slots[i] >= (uint32_t)s->max_rounds) return HBLS_ERR_BADINPUT;
+    size_t bm = (s->c->n + 7) / 8;
+    HIP_OK(hipMemcpy(s->ck_slots, slots, k * 4, hipMemcpyHostToDevice));
+    /* snapshot on stream 0 (after the ticks) into the persistent ck_* bufs */
+    hipLaunchKernelGGL(k_stream_gather_check, dim3(k), dim3(64), 0, 0,
+                       s->d_bitmap, s->nwords, (int)bm, s->d_hm, s->d_hm_ok,
+                       s->d_agg, s->ck_slots, k,
+                       s->ck_bm, s->ck_hm, s->ck_hok, s->ck_saff, s->ck_sflags);
+    hipEvent_t ev;
+    HIP_OK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    HIP_OK(hipEventRecord(ev, 0));
+    HIP_OK(hipStreamWaitEvent(s->check_stream, ev, 0));
+    (void)hipEventDestroy(ev);
+    launch_mask_aggregate(s->c->d_table, (int)s->c->n, s->ck_bm, (int)bm,
+                          s->c->d_full_sum, s->ck_agg, k,
+                          s->c->d_wtab, s->c->d_winf, s->check_stream);
+    int nbc = (k + CV_ITEMS - 1) / CV_ITEMS;
+    hipLaunchKernelGGL(k_verify_coop<16>, dim3(nbc), dim3(64), 0, s->check_stream,
+                       s->ck_agg, s->ck_hm, s->ck_saff,
+                       s->ck_sflags, s->ck_hok, s->ck_res, k);
+    HIP_OK(hipGetLastError());
+    s->ck_pending = k;
+    return HBLS_OK;
+}
+
+/* wait for the in-flight check and return its per-round verdicts.
+ * ok must hold at least the submitted k entries; returns that k. */
+extern "C" int hbls_stream_check_poll(hbls_stream *s, int32_t *ok) {
+    int rc = require_gpu();
+    if (rc != HBLS_OK) return rc;
+    if (!s->ck_pending) return 0;
+    HIP_OK(hipStreamSynchronize(s->check_stream));
+    int k = s->ck_pending;
+    HIP_OK(hipMemcpy(ok, s->ck_res, k * 4, hipMemcpyDeviceToHost));
+    s->ck_pending = 0;
+    return k;
 }
 
 /* export one round's state: byte bitmap + serialized aggregate (the shape

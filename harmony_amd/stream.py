@@ -150,13 +150,23 @@ class MultiStreamVerifier:
             self.window_checks += len(due)
             for r in due:
                 self._since_check[r] = 0
-            if not all(self.stream.check(due)):
-                raise RuntimeError("streaming aggregate diverged from mask")
+            # async: collect the PREVIOUS window check's verdicts, then
+            # submit this one on the side stream so the pairing chain
+            # overlaps the next ticks (divergence detection trails by one
+            # window, like the reference's eventually-checked aggregates)
+            self._collect_window_check()
+            self.stream.check_submit(due)
         return [r if r == 1 else 0 for r in res]
+
+    def _collect_window_check(self):
+        verdicts = self.stream.check_poll()
+        if verdicts and not all(verdicts.values()):
+            raise RuntimeError("streaming aggregate diverged from mask")
 
     def reset_rounds(self, payloads):
         """open a fresh set of rounds in place (new block heights): hashes
         the payloads on device and clears bitmaps + aggregates, one call."""
+        self._collect_window_check()   # verdicts belong to the old rounds
         if len(payloads) != len(self.payloads) or \
                 any(len(p) != self.mlen for p in payloads):
             raise ValueError("reset_rounds: shape mismatch")
@@ -170,4 +180,5 @@ class MultiStreamVerifier:
 
     def final_check_all(self) -> bool:
         """one batched aggregate-verify across every round in flight"""
+        self._collect_window_check()
         return all(self.stream.check(list(range(len(self.rounds)))))

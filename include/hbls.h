@@ -194,6 +194,12 @@ int hbls_stream_process(hbls_stream_t *s, const uint32_t *key_idx,
 /* pairing-check rounds' resident aggregates vs their masks
  * (validator.go:224-228); ok[i] = 1/0 */
 int hbls_stream_check(hbls_stream_t *s, const uint32_t *slots, int k, int32_t *ok);
+/* async form: snapshot the rounds' state (ordered after prior ticks) and run
+ * the latency-bound check chain on a dedicated HIP stream, overlapping
+ * subsequent ticks; one check in flight at a time.  poll waits and returns
+ * the submitted k (0 if none pending). */
+int hbls_stream_check_submit(hbls_stream_t *s, const uint32_t *slots, int k);
+int hbls_stream_check_poll(hbls_stream_t *s, int32_t *ok);
 /* export a round's bitmap + serialized aggregate
  * (consensus_service.go:305-322 commitSigAndBitmap shape) */
 int hbls_stream_get(hbls_stream_t *s, uint32_t slot, uint8_t *bitmap_out,

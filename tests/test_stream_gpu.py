@@ -179,3 +179,34 @@ def test_stream_ragged_committee(oracle_lib):
         assert len(bm) == 5
         assert bm == bytes([0x01, 0x00, 0x00, 0x80, 0x01])   # keys 0, 31, 32
         assert oc.agg_verify(bm, agg, payloads[r]) is True
+
+
+def test_stream_async_check(oracle_lib):
+    """hbls_stream_check_submit/poll: the snapshot-based side-stream check
+    returns the same verdicts as the synchronous check while later ticks
+    proceed; double-submit rejects; empty poll is a no-op."""
+    from harmony_amd import core
+    n = 32
+    sks = [pr.fr_serialize(pr.synth_sk(i)) for i in range(n)]
+    pks = core.batch_pk_from_sk(b"".join(sks), n)
+    st = core.Stream(core.Committee(pks, n), 2)
+    payloads = [pr.construct_commit_payload(r, pr.synth_msg(500 + r), r)
+                for r in range(2)]
+    st.set_rounds([0, 1], b"".join(payloads), len(payloads[0]))
+    votes = [(r, i) for r in range(2) for i in range(0, n, 2)]
+    sigs = b"".join(oracle_lib.sign_hash(sks[i], payloads[r]) for r, i in votes)
+    assert all(x == 1 for x in st.process([v[1] for v in votes],
+                                          [v[0] for v in votes], sigs))
+    assert st.check_poll() == {}                 # nothing pending
+    st.check_submit([0, 1])
+    with pytest.raises(ValueError):
+        st.check_submit([0])                     # one in flight at a time
+    # overlap: more ticks while the check runs on the side stream
+    votes2 = [(r, i) for r in range(2) for i in range(1, n, 2)]
+    sigs2 = b"".join(oracle_lib.sign_hash(sks[i], payloads[r]) for r, i in votes2)
+    assert all(x == 1 for x in st.process([v[1] for v in votes2],
+                                          [v[0] for v in votes2], sigs2))
+    # the snapshot verdict covers the state at submit time (half the votes)
+    assert st.check_poll() == {0: True, 1: True}
+    # synchronous check of the final state agrees
+    assert st.check([0, 1]) == [True, True]
