@@ -182,3 +182,17 @@ class MultiStreamVerifier:
         """one batched aggregate-verify across every round in flight"""
         self._collect_window_check()
         return all(self.stream.check(list(range(len(self.rounds)))))
+
+    def final_check_submit(self):
+        """async form of final_check_all: snapshots every round's state now
+        (safe across a following reset_rounds — the snapshot is taken on
+        the tick stream before the reset) and verifies on the side stream;
+        collect the verdict with final_check_collect.  The pipelined shape:
+        submit at the end of one block batch, collect at the start of the
+        next — every aggregate still gets checked, one batch later."""
+        self._collect_window_check()
+        self.stream.check_submit(list(range(len(self.rounds))))
+
+    def final_check_collect(self) -> bool:
+        verdicts = self.stream.check_poll()
+        return all(verdicts.values()) if verdicts else True
